@@ -1,0 +1,122 @@
+"""Single config object consumed by every entry point.
+
+The reference scattered ~8 argparse flags across six scripts and shipped a dead
+``utils/config.py`` (reference utils/config.py:1-10 — imported by nothing). Here
+the dataclass is the one source of truth; the reference's intended-but-unwired
+knobs (save_epoch, tensorboard/metrics dir) are real.
+"""
+from __future__ import annotations
+
+import argparse
+import dataclasses
+import os
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+# CIFAR-100 channel statistics (reference utils/dataset.py:8 hard-codes these).
+CIFAR100_MEAN = (0.5070751592371323, 0.48654887331495095, 0.4409178433670343)
+CIFAR100_STD = (0.2673342858792401, 0.2564384629170883, 0.27615047132568404)
+
+
+@dataclass
+class TrainConfig:
+    # model / task
+    arch: str = "resnet18"            # resnet18 | resnet34 | resnet50
+    num_classes: int = 100
+    image_size: int = 32
+
+    # optimisation (reference defaults: distributed.py:18-25,63-64)
+    batch_size: int = 256             # GLOBAL batch, divided by world size
+    epochs: int = 200
+    lr: float = 0.1
+    momentum: float = 0.9
+    weight_decay: float = 1e-4
+    lr_milestones: List[int] = field(default_factory=lambda: [60, 120, 160])
+    lr_gamma: float = 0.2
+    grad_accu_steps: int = 1          # micro-batching; collectives elided on non-final micro-steps
+
+    # precision: "fp32" | "bf16" | "fp16" (fp16 uses the dynamic loss scaler)
+    amp: str = "fp32"
+
+    # distributed
+    ip: str = "127.0.0.1"
+    port: int = 23456
+    backend: Optional[str] = None     # None -> nccl(RCCL) when a GPU is visible, else gloo
+    sync_bn: bool = True              # reference enables SyncBN in every DDP entry
+    bucket_cap_mb: int = 25           # DDP gradient bucket size (tuned for xGMI on HW)
+    use_flat_ddp: bool = True         # MI355X-native flat-bucket reducer (graph-capturable)
+
+    # data
+    data_root: str = "./data"
+    synthetic: bool = True            # no network in this environment; real CIFAR if present
+    num_workers: int = 4
+    pin_memory: bool = True
+    seed: int = 0
+
+    # native kernels
+    native_ops: bool = True           # HIP kernels when extension present; fail loudly on GPU if absent
+    channels_last: bool = False
+    hip_graph: bool = False           # capture the train step in a hipGraph
+
+    # logging / checkpointing (makes reference utils/config.py's dead knobs real)
+    log_interval: int = 10            # steps between loss all-reduce + print (reference did it every step)
+    metrics_dir: str = "runs"
+    ckpt_dir: str = "ckpts"
+    save_epoch: int = 15
+    eval_every_epoch: bool = True
+    resume: str = ""
+
+    def per_rank_batch(self, world_size: int) -> int:
+        return max(1, self.batch_size // max(1, world_size))
+
+    def replace(self, **kw) -> "TrainConfig":
+        return dataclasses.replace(self, **kw)
+
+
+def add_common_args(p: argparse.ArgumentParser) -> argparse.ArgumentParser:
+    """Reference-compatible flag surface (reference distributed.py:18-25)."""
+    p.add_argument("--seed", default=0, type=int, help="seed for initializing training")
+    p.add_argument("--batch_size", "--batch-size", default=256, type=int,
+                   help="global batch size across all GPUs")
+    p.add_argument("--epochs", default=200, type=int)
+    p.add_argument("--lr", "--learning-rate", default=0.1, type=float)
+    p.add_argument("--ip", default="127.0.0.1", type=str)
+    p.add_argument("--port", default=23456, type=int)
+    p.add_argument("--arch", default="resnet18", type=str,
+                   choices=["resnet18", "resnet34", "resnet50"])
+    p.add_argument("--amp", default=None, type=str, choices=[None, "fp32", "bf16", "fp16"])
+    p.add_argument("--no-sync-bn", action="store_true")
+    p.add_argument("--num_workers", default=4, type=int)
+    p.add_argument("--log_interval", default=10, type=int)
+    p.add_argument("--data_root", default="./data", type=str)
+    p.add_argument("--ckpt_dir", default="ckpts", type=str)
+    p.add_argument("--metrics_dir", default="runs", type=str)
+    p.add_argument("--save_epoch", default=15, type=int)
+    p.add_argument("--resume", default="", type=str)
+    p.add_argument("--synthetic", action="store_true", default=None,
+                   help="force synthetic CIFAR-shaped data (default: auto)")
+    return p
+
+
+def config_from_args(args: argparse.Namespace, **overrides) -> TrainConfig:
+    cfg = TrainConfig()
+    mapping = dict(
+        seed="seed", batch_size="batch_size", epochs="epochs", lr="lr", ip="ip",
+        port="port", arch="arch", num_workers="num_workers", log_interval="log_interval",
+        data_root="data_root", ckpt_dir="ckpt_dir", metrics_dir="metrics_dir",
+        save_epoch="save_epoch", resume="resume",
+    )
+    kw = {}
+    for argname, cfgname in mapping.items():
+        if hasattr(args, argname) and getattr(args, argname) is not None:
+            kw[cfgname] = getattr(args, argname)
+    if getattr(args, "amp", None):
+        kw["amp"] = args.amp
+    if getattr(args, "no_sync_bn", False):
+        kw["sync_bn"] = False
+    if getattr(args, "synthetic", None) is not None:
+        kw["synthetic"] = bool(args.synthetic)
+    if getattr(args, "grad_accu_steps", None):
+        kw["grad_accu_steps"] = args.grad_accu_steps
+    kw.update(overrides)
+    return TrainConfig(**kw)

@@ -1,0 +1,138 @@
+"""CIFAR-variant ResNet-18/34/50 (capability parity with reference utils/model.py).
+
+Same architecture family as the reference model zoo (3x3 stem, no maxpool,
+32x32 inputs, four stages, num_classes=100 — reference utils/model.py:61-127)
+but written for the MI355X execution path: every BatchNorm+ReLU pair and the
+residual BN+add+ReLU join dispatch to fused HIP kernels through
+ops.bn_relu / ops.bn_add_relu (one kernel instead of 2-3 ATen launches), with
+the module tree kept as standard nn.Conv2d/nn.BatchNorm2d so checkpoints,
+SyncBN conversion and torch-eager parity A/B all operate on the same state.
+Convolutions run through nn.Conv2d (MIOpen) or the native implicit-GEMM HIP
+kernel when enabled (ops.conv, added by the kernel campaign).
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+
+from ..ops import bn_relu, bn_add_relu
+
+
+def _conv3x3(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
+    return nn.Conv2d(cin, cout, kernel_size=3, stride=stride, padding=1, bias=False)
+
+
+def _conv1x1(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
+    return nn.Conv2d(cin, cout, kernel_size=1, stride=stride, bias=False)
+
+
+class BasicBlock(nn.Module):
+    """Two 3x3 convs; expansion 1 (reference utils/model.py:3-28)."""
+
+    expansion = 1
+
+    def __init__(self, cin: int, cout: int, stride: int = 1):
+        super().__init__()
+        self.conv1 = _conv3x3(cin, cout, stride)
+        self.bn1 = nn.BatchNorm2d(cout)
+        self.conv2 = _conv3x3(cout, cout)
+        self.bn2 = nn.BatchNorm2d(cout)
+        self.downsample = None
+        if stride != 1 or cin != cout * self.expansion:
+            self.downsample = nn.Sequential(
+                _conv1x1(cin, cout * self.expansion, stride),
+                nn.BatchNorm2d(cout * self.expansion),
+            )
+
+    def forward(self, x):
+        identity = x
+        out = bn_relu(self.conv1(x), self.bn1)
+        out = self.conv2(out)
+        if self.downsample is not None:
+            identity = bn_relu(self.downsample[0](x), self.downsample[1], relu=False)
+        return bn_add_relu(out, identity, self.bn2)
+
+
+class Bottleneck(nn.Module):
+    """1x1 -> 3x3 -> 1x1; expansion 4 (reference utils/model.py:32-59, minus
+    its stray debug print at :37)."""
+
+    expansion = 4
+
+    def __init__(self, cin: int, cout: int, stride: int = 1):
+        super().__init__()
+        self.conv1 = _conv1x1(cin, cout)
+        self.bn1 = nn.BatchNorm2d(cout)
+        self.conv2 = _conv3x3(cout, cout, stride)
+        self.bn2 = nn.BatchNorm2d(cout)
+        self.conv3 = _conv1x1(cout, cout * self.expansion)
+        self.bn3 = nn.BatchNorm2d(cout * self.expansion)
+        self.downsample = None
+        if stride != 1 or cin != cout * self.expansion:
+            self.downsample = nn.Sequential(
+                _conv1x1(cin, cout * self.expansion, stride),
+                nn.BatchNorm2d(cout * self.expansion),
+            )
+
+    def forward(self, x):
+        identity = x
+        out = bn_relu(self.conv1(x), self.bn1)
+        out = bn_relu(self.conv2(out), self.bn2)
+        out = self.conv3(out)
+        if self.downsample is not None:
+            identity = bn_relu(self.downsample[0](x), self.downsample[1], relu=False)
+        return bn_add_relu(out, identity, self.bn3)
+
+
+class ResNet(nn.Module):
+    def __init__(self, block, num_blocks, num_classes: int = 100):
+        super().__init__()
+        self.in_channels = 64
+        # CIFAR stem: single 3x3, stride 1, no maxpool (reference utils/model.py:66-70)
+        self.conv1 = _conv3x3(3, 64)
+        self.bn1 = nn.BatchNorm2d(64)
+        self.layer1 = self._make_layer(block, 64, num_blocks[0], 1)
+        self.layer2 = self._make_layer(block, 128, num_blocks[1], 2)
+        self.layer3 = self._make_layer(block, 256, num_blocks[2], 2)
+        self.layer4 = self._make_layer(block, 512, num_blocks[3], 2)
+        self.avgpool = nn.AdaptiveAvgPool2d((1, 1))
+        self.fc = nn.Linear(512 * block.expansion, num_classes)
+
+    def _make_layer(self, block, cout, blocks, stride):
+        strides = [stride] + [1] * (blocks - 1)
+        layers = []
+        for s in strides:
+            layers.append(block(self.in_channels, cout, s))
+            self.in_channels = cout * block.expansion
+        return nn.Sequential(*layers)
+
+    def forward(self, x):
+        out = bn_relu(self.conv1(x), self.bn1)
+        out = self.layer1(out)
+        out = self.layer2(out)
+        out = self.layer3(out)
+        out = self.layer4(out)
+        out = self.avgpool(out)
+        out = torch.flatten(out, 1)
+        return self.fc(out)
+
+
+def resnet18(num_classes: int = 100) -> ResNet:
+    return ResNet(BasicBlock, [2, 2, 2, 2], num_classes)
+
+
+def resnet34(num_classes: int = 100) -> ResNet:
+    return ResNet(BasicBlock, [3, 4, 6, 3], num_classes)
+
+
+def resnet50(num_classes: int = 100) -> ResNet:
+    return ResNet(Bottleneck, [3, 4, 6, 3], num_classes)
+
+
+_FACTORIES = {"resnet18": resnet18, "resnet34": resnet34, "resnet50": resnet50}
+
+
+def build_model(arch: str, num_classes: int = 100) -> ResNet:
+    if arch not in _FACTORIES:
+        raise ValueError(f"unknown arch {arch!r}; choose from {sorted(_FACTORIES)}")
+    return _FACTORIES[arch](num_classes)
