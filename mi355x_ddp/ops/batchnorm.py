@@ -61,9 +61,10 @@ def _bwd_apply(dy, x, weight, mean, invstd, sum_dy, sum_dy_xhat, count, y, relu,
                training, need_dresidual):
     """dx (+ dresidual) for the fused BN(+add)(+relu)."""
     if _backend.native_enabled(x):
-        return _backend.C().bn_bwd(dy, x, weight, mean, invstd, sum_dy,
-                                   sum_dy_xhat, float(count), y, relu, training,
-                                   need_dresidual)
+        out = _backend.C().bn_bwd(dy, x, weight.float().contiguous(), mean,
+                                  invstd, sum_dy, sum_dy_xhat, float(count), y,
+                                  relu, training, need_dresidual)
+        return out[0], (out[1] if need_dresidual else None)
     g = dy.float()
     if relu:
         g = g * (y > 0).float()
@@ -122,6 +123,11 @@ class _FusedBNFunction(torch.autograd.Function):
         dy = dy.contiguous()
         sum_dy, sum_dy_xhat = _bwd_reduce(dy, x, mean, invstd, y, ctx.relu)
         count = ctx.count
+        # dgamma/dbeta are the LOCAL sums — the DP gradient all-reduce averages
+        # them like every other parameter grad. The cross-rank-summed versions
+        # are only for dx (whose formula needs the GLOBAL batch statistics).
+        dgamma = sum_dy_xhat.to(weight.dtype)
+        dbeta = sum_dy.to(weight.dtype)
         if ctx.training and ctx.process_group is not None and dist.is_initialized() \
                 and dist.get_world_size(ctx.process_group) > 1:
             C = sum_dy.numel()
@@ -131,8 +137,6 @@ class _FusedBNFunction(torch.autograd.Function):
         dx, dresidual = _bwd_apply(dy, x, weight, mean, invstd, sum_dy,
                                    sum_dy_xhat, count, y, ctx.relu,
                                    ctx.training, ctx.has_residual)
-        dgamma = sum_dy_xhat.to(weight.dtype)
-        dbeta = sum_dy.to(weight.dtype)
         return (dx, dresidual, dgamma, dbeta, None, None, None, None, None, None, None)
 
 

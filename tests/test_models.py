@@ -1,0 +1,48 @@
+import pytest
+import torch
+
+from mi355x_ddp.models import build_model, resnet18, resnet34, resnet50
+
+
+@pytest.mark.parametrize("arch,expansion", [("resnet18", 1), ("resnet34", 1),
+                                            ("resnet50", 4)])
+def test_forward_shapes(arch, expansion):
+    model = build_model(arch, num_classes=100)
+    x = torch.randn(4, 3, 32, 32)
+    y = model(x)
+    assert y.shape == (4, 100)
+    assert model.fc.in_features == 512 * expansion
+
+
+def test_resnet18_param_count():
+    # CIFAR ResNet18 w/ 100 classes: ~11.22 M params
+    n = sum(p.numel() for p in resnet18().parameters())
+    assert 11.0e6 < n < 11.5e6
+
+
+def test_backward_flows():
+    model = resnet18()
+    x = torch.randn(2, 3, 32, 32)
+    loss = model(x).sum()
+    loss.backward()
+    grads = [p.grad for p in model.parameters()]
+    assert all(g is not None for g in grads)
+    assert all(torch.isfinite(g).all() for g in grads)
+
+
+def test_eval_mode_uses_running_stats():
+    model = resnet18()
+    model.eval()
+    x = torch.randn(2, 3, 32, 32)
+    y1 = model(x)
+    y2 = model(x)
+    assert torch.allclose(y1, y2)
+
+
+def test_state_dict_roundtrip():
+    m1 = resnet34()
+    m2 = resnet34()
+    m2.load_state_dict(m1.state_dict())
+    x = torch.randn(2, 3, 32, 32)
+    m1.eval(), m2.eval()
+    assert torch.allclose(m1(x), m2(x))

@@ -1,0 +1,133 @@
+"""Fused-op front-ends vs plain torch references (CPU fallback path).
+
+The same autograd.Functions drive the HIP kernels on GPU; here the math and
+autograd wiring are verified against eager torch compositions.
+"""
+import pytest
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from mi355x_ddp.core.amp import DynamicLossScaler
+from mi355x_ddp.ops import FusedSGD, bn_add_relu, bn_relu, softmax_cross_entropy
+
+
+def _clone_bn(bn):
+    ref = nn.BatchNorm2d(bn.num_features, eps=bn.eps, momentum=bn.momentum)
+    ref.load_state_dict(bn.state_dict())
+    return ref
+
+
+@pytest.mark.parametrize("training", [True, False])
+@pytest.mark.parametrize("relu", [True, False])
+def test_bn_relu_matches_torch(training, relu):
+    torch.manual_seed(0)
+    bn = nn.BatchNorm2d(8)
+    ref_bn = _clone_bn(bn)
+    bn.train(training), ref_bn.train(training)
+
+    x1 = torch.randn(4, 8, 5, 5, requires_grad=True)
+    x2 = x1.detach().clone().requires_grad_(True)
+
+    y = bn_relu(x1, bn, relu=relu)
+    y_ref = ref_bn(x2)
+    if relu:
+        y_ref = F.relu(y_ref)
+    assert torch.allclose(y, y_ref, atol=1e-5), (y - y_ref).abs().max()
+
+    g = torch.randn_like(y)
+    y.backward(g)
+    y_ref.backward(g)
+    assert torch.allclose(x1.grad, x2.grad, atol=1e-5)
+    assert torch.allclose(bn.weight.grad, ref_bn.weight.grad, atol=1e-4)
+    assert torch.allclose(bn.bias.grad, ref_bn.bias.grad, atol=1e-4)
+    if training:
+        assert torch.allclose(bn.running_mean, ref_bn.running_mean, atol=1e-6)
+        assert torch.allclose(bn.running_var, ref_bn.running_var, atol=1e-6)
+
+
+def test_bn_add_relu_matches_torch():
+    torch.manual_seed(1)
+    bn = nn.BatchNorm2d(6)
+    ref_bn = _clone_bn(bn)
+    x1 = torch.randn(3, 6, 4, 4, requires_grad=True)
+    r1 = torch.randn(3, 6, 4, 4, requires_grad=True)
+    x2 = x1.detach().clone().requires_grad_(True)
+    r2 = r1.detach().clone().requires_grad_(True)
+
+    y = bn_add_relu(x1, r1, bn)
+    y_ref = F.relu(ref_bn(x2) + r2)
+    assert torch.allclose(y, y_ref, atol=1e-5)
+
+    g = torch.randn_like(y)
+    y.backward(g)
+    y_ref.backward(g)
+    assert torch.allclose(x1.grad, x2.grad, atol=1e-5)
+    assert torch.allclose(r1.grad, r2.grad, atol=1e-5)
+
+
+def test_softmax_cross_entropy_matches_torch():
+    torch.manual_seed(2)
+    logits1 = torch.randn(16, 100, requires_grad=True)
+    logits2 = logits1.detach().clone().requires_grad_(True)
+    target = torch.randint(0, 100, (16,))
+
+    loss = softmax_cross_entropy(logits1, target)
+    loss_ref = F.cross_entropy(logits2, target)
+    assert torch.allclose(loss, loss_ref, atol=1e-6)
+
+    loss.backward()
+    loss_ref.backward()
+    assert torch.allclose(logits1.grad, logits2.grad, atol=1e-6)
+
+
+def test_fused_sgd_matches_torch_sgd():
+    torch.manual_seed(3)
+    m1 = nn.Linear(10, 10)
+    m2 = nn.Linear(10, 10)
+    m2.load_state_dict(m1.state_dict())
+    o1 = FusedSGD(m1.parameters(), lr=0.1, momentum=0.9, weight_decay=1e-4)
+    o2 = torch.optim.SGD(m2.parameters(), lr=0.1, momentum=0.9, weight_decay=1e-4)
+    x = torch.randn(4, 10)
+    for _ in range(5):
+        for m, o in ((m1, o1), (m2, o2)):
+            o.zero_grad()
+            m(x).pow(2).mean().backward()
+            o.step()
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p1, p2, atol=1e-6)
+
+
+def test_loss_scaler_skips_on_overflow_and_backs_off():
+    scaler = DynamicLossScaler(init_scale=2.0 ** 8, growth_interval=2)
+    m = nn.Linear(4, 4)
+    opt = FusedSGD(m.parameters(), lr=0.1)
+    before = [p.detach().clone() for p in m.parameters()]
+
+    # poison a grad with inf -> step must be skipped, scale halved
+    m(torch.randn(2, 4)).sum().backward()
+    with torch.no_grad():
+        next(m.parameters()).grad[0, 0] = float("inf")
+    scaler.unscale_([p.grad for p in m.parameters()])
+    assert scaler.found_inf
+    stepped = scaler.step(opt)
+    assert not stepped
+    assert scaler.scale == 2.0 ** 7
+    for p, b in zip(m.parameters(), before):
+        assert torch.equal(p, b)
+
+    # clean steps grow the scale after growth_interval
+    for _ in range(2):
+        opt.zero_grad()
+        m(torch.randn(2, 4)).sum().backward()
+        scaler.unscale_([p.grad for p in m.parameters()])
+        assert not scaler.found_inf
+        assert scaler.step(opt)
+    assert scaler.scale == 2.0 ** 8
+
+
+def test_scaler_unscale_divides():
+    scaler = DynamicLossScaler(init_scale=4.0)
+    g = torch.full((8,), 4.0)
+    scaler.unscale_([g])
+    assert torch.allclose(g, torch.ones(8))

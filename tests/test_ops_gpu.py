@@ -1,0 +1,197 @@
+"""HIP kernel numerics vs plain PyTorch fp32 references (runs on MI355X).
+
+Every native op is compared against an eager fp32 torch composition on the
+same data; bf16/fp16 variants use dtype-appropriate tolerance bands.
+"""
+import os
+
+import pytest
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda"
+SHAPES = [  # every ResNet18/50 CIFAR stage shape
+    (8, 64, 32, 32),
+    (8, 128, 16, 16),
+    (8, 256, 8, 8),
+    (8, 512, 4, 4),
+]
+
+
+def tol(dtype):
+    return dict(atol=1e-4, rtol=1e-4) if dtype == torch.float32 else \
+        dict(atol=5e-2, rtol=5e-2)
+
+
+@pytest.fixture(autouse=True)
+def _native_required():
+    from mi355x_ddp.ops import _backend
+    assert _backend.extension_available(), "HIP extension must be built"
+    yield
+
+
+@pytest.mark.parametrize("shape", SHAPES)
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_bn_stats_kernel(shape, dtype):
+    from mi355x_ddp.ops import _backend
+    x = torch.randn(*shape, device=DEV, dtype=dtype)
+    s, sq = _backend.C().bn_stats(x)
+    xf = x.float()
+    assert torch.allclose(s, xf.sum((0, 2, 3)), **tol(dtype))
+    assert torch.allclose(sq, (xf * xf).sum((0, 2, 3)), **tol(dtype))
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("relu", [False, True])
+@pytest.mark.parametrize("residual", [False, True])
+def test_bn_fwd_kernel(dtype, relu, residual):
+    from mi355x_ddp.ops import _backend
+    N, C, H, W = 4, 32, 8, 8
+    x = torch.randn(N, C, H, W, device=DEV, dtype=dtype)
+    res = torch.randn_like(x) if residual else torch.empty(0, device=DEV, dtype=dtype)
+    w = torch.rand(C, device=DEV) + 0.5
+    b = torch.randn(C, device=DEV)
+    mean = torch.randn(C, device=DEV)
+    invstd = torch.rand(C, device=DEV) + 0.5
+    y = _backend.C().bn_fwd(x, w, b, mean, invstd, relu, res)
+    sh = (1, -1, 1, 1)
+    ref = (x.float() - mean.view(sh)) * invstd.view(sh) * w.view(sh) + b.view(sh)
+    if residual:
+        ref = ref + res.float()
+    if relu:
+        ref = F.relu(ref)
+    assert torch.allclose(y.float(), ref.to(dtype).float(), **tol(dtype))
+
+
+@pytest.mark.parametrize("training", [True, False])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_fused_bn_relu_autograd_vs_torch(training, dtype):
+    from mi355x_ddp.ops import bn_relu
+    torch.manual_seed(0)
+    bn = nn.BatchNorm2d(64).to(DEV)
+    ref_bn = nn.BatchNorm2d(64).to(DEV)
+    ref_bn.load_state_dict(bn.state_dict())
+    bn.train(training), ref_bn.train(training)
+
+    x1 = torch.randn(8, 64, 16, 16, device=DEV, dtype=dtype, requires_grad=True)
+    x2 = x1.detach().float().requires_grad_(True)
+
+    y = bn_relu(x1, bn)
+    y_ref = F.relu(ref_bn(x2))
+    assert torch.allclose(y.float(), y_ref, **tol(dtype))
+
+    g = torch.randn_like(y_ref)
+    y.backward(g.to(dtype))
+    y_ref.backward(g)
+    assert torch.allclose(x1.grad.float(), x2.grad, **tol(dtype))
+    assert torch.allclose(bn.weight.grad, ref_bn.weight.grad,
+                          atol=1e-2, rtol=1e-2)
+    assert torch.allclose(bn.bias.grad, ref_bn.bias.grad,
+                          atol=1e-2, rtol=1e-2)
+    if training:
+        assert torch.allclose(bn.running_mean, ref_bn.running_mean, atol=1e-3)
+        assert torch.allclose(bn.running_var, ref_bn.running_var, atol=1e-3)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_fused_bn_add_relu_autograd_vs_torch(dtype):
+    from mi355x_ddp.ops import bn_add_relu
+    torch.manual_seed(1)
+    bn = nn.BatchNorm2d(128).to(DEV)
+    ref_bn = nn.BatchNorm2d(128).to(DEV)
+    ref_bn.load_state_dict(bn.state_dict())
+
+    x1 = torch.randn(4, 128, 8, 8, device=DEV, dtype=dtype, requires_grad=True)
+    r1 = torch.randn(4, 128, 8, 8, device=DEV, dtype=dtype, requires_grad=True)
+    x2 = x1.detach().float().requires_grad_(True)
+    r2 = r1.detach().float().requires_grad_(True)
+
+    y = bn_add_relu(x1, r1, bn)
+    y_ref = F.relu(ref_bn(x2) + r2)
+    assert torch.allclose(y.float(), y_ref, **tol(dtype))
+    g = torch.randn_like(y_ref)
+    y.backward(g.to(dtype))
+    y_ref.backward(g)
+    assert torch.allclose(x1.grad.float(), x2.grad, **tol(dtype))
+    assert torch.allclose(r1.grad.float(), r2.grad, **tol(dtype))
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("nc", [(16, 100), (256, 100), (64, 1000)])
+def test_xent_kernel_vs_torch(dtype, nc):
+    from mi355x_ddp.ops import softmax_cross_entropy
+    N, C = nc
+    torch.manual_seed(2)
+    logits1 = torch.randn(N, C, device=DEV, dtype=dtype, requires_grad=True)
+    logits2 = logits1.detach().float().requires_grad_(True)
+    target = torch.randint(0, C, (N,), device=DEV)
+
+    loss = softmax_cross_entropy(logits1, target)
+    loss_ref = F.cross_entropy(logits2, target)
+    assert torch.allclose(loss.float(), loss_ref, atol=1e-2, rtol=1e-3)
+    loss.backward()
+    loss_ref.backward()
+    assert torch.allclose(logits1.grad.float(), logits2.grad, **tol(dtype))
+
+
+def test_multi_tensor_sgd_vs_torch():
+    from mi355x_ddp.ops import _backend
+    torch.manual_seed(3)
+    # 70 tensors of mixed sizes -> exercises multi-launch chunking (cap 32)
+    sizes = [(3,), (64,), (64, 3, 3, 3), (512, 512, 3, 3), (100, 512), (1000000,)] * 12
+    params = [torch.randn(*s, device=DEV) for s in sizes[:70]]
+    grads = [torch.randn_like(p) for p in params]
+    bufs = [torch.randn_like(p) for p in params]
+    ref_p = [p.clone() for p in params]
+    ref_m = [m.clone() for m in bufs]
+
+    lr, mu, wd = 0.1, 0.9, 1e-4
+    _backend.C().multi_tensor_sgd(params, grads, bufs, lr, mu, wd)
+    for p, g, m in zip(ref_p, grads, ref_m):
+        d = g.add(p, alpha=wd)
+        m.mul_(mu).add_(d)
+        p.add_(m, alpha=-lr)
+    for p, rp in zip(params, ref_p):
+        assert torch.allclose(p, rp, atol=1e-6)
+    for m, rm in zip(bufs, ref_m):
+        assert torch.allclose(m, rm, atol=1e-6)
+
+
+def test_multi_tensor_unscale_detects_inf_and_nan():
+    from mi355x_ddp.ops import _backend
+    g1 = torch.full((1 << 18,), 4.0, device=DEV)
+    g2 = torch.full((123,), 2.0, device=DEV)
+    found = _backend.C().multi_tensor_unscale([g1, g2], 0.5)
+    assert found.item() == 0
+    assert torch.allclose(g1, torch.full_like(g1, 2.0))
+    assert torch.allclose(g2, torch.full_like(g2, 1.0))
+
+    g1[777] = float("inf")
+    found = _backend.C().multi_tensor_unscale([g1, g2], 1.0)
+    assert found.item() == 1
+    g1[777] = 0.0
+    g2[5] = float("nan")
+    found = _backend.C().multi_tensor_unscale([g1, g2], 1.0)
+    assert found.item() == 1
+
+
+def test_fused_sgd_optimizer_gpu_matches_torch():
+    from mi355x_ddp.ops import FusedSGD
+    torch.manual_seed(4)
+    m1 = nn.Sequential(nn.Conv2d(3, 8, 3), nn.Flatten(), nn.LazyLinear(10)).to(DEV)
+    m1(torch.randn(2, 3, 8, 8, device=DEV))  # materialise lazy
+    import copy
+    m2 = copy.deepcopy(m1)
+    o1 = FusedSGD(m1.parameters(), lr=0.05, momentum=0.9, weight_decay=1e-4)
+    o2 = torch.optim.SGD(m2.parameters(), lr=0.05, momentum=0.9, weight_decay=1e-4)
+    x = torch.randn(4, 3, 8, 8, device=DEV)
+    for _ in range(3):
+        for m, o in ((m1, o1), (m2, o2)):
+            o.zero_grad()
+            m(x).pow(2).mean().backward()
+            o.step()
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p1, p2, atol=1e-5)
