@@ -91,8 +91,21 @@ def main():
             img, lbl = img.pin_memory(), lbl.pin_memory()
         batches.append((img, lbl))
 
+    graph_step = None
+    if args.hip_graph:
+        assert device.type == "cuda", "--hip-graph needs a GPU"
+        assert args.mode == "flat" and accu == 1 and args.amp != "fp16"
+        from mi355x_ddp.core.graphs import GraphedTrainStep
+        model.train()
+        graph_step = GraphedTrainStep(model, criterion, optimizer, cfg, device,
+                                      batch=per_rank, image_size=args.image_size)
+
     def one_step(i: int):
         img_h, lbl_h = batches[i % n_prefab]
+        if graph_step is not None:
+            # static-buffer copy_ pulls straight from pinned host memory
+            graph_step.run(img_h, lbl_h)
+            return
         images = img_h.to(device, non_blocking=True)
         labels = lbl_h.to(device, non_blocking=True)
         if args.channels_last:

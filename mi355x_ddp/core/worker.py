@@ -49,7 +49,8 @@ def build_training(cfg: TrainConfig, device: torch.device, world_size: int,
     if distributed and world_size > 1 and cfg.sync_bn:
         model = MI355SyncBatchNorm.convert_sync_batchnorm(model)
     if distributed and world_size >= 1 and wrap == "flat":
-        model = FlatDDP(model, bucket_cap_mb=cfg.bucket_cap_mb)
+        model = FlatDDP(model, bucket_cap_mb=cfg.bucket_cap_mb,
+                        overlap=not cfg.hip_graph)
     elif distributed and wrap == "torch":
         model = wrap_torch_ddp(model,
                                device.index if device.type == "cuda" else None,

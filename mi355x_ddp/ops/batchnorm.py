@@ -89,13 +89,16 @@ class _FusedBNFunction(torch.autograd.Function):
         N, C, H, W = x.shape
         if training:
             s, sq = _stats(x)
-            count = torch.full((1,), float(N * H * W), device=x.device, dtype=torch.float32)
+            # count is computed arithmetically (equal per-rank batches are
+            # guaranteed by drop_last sharding) — no .item() host sync per BN
+            # layer, which also keeps the op hipGraph-capturable.
+            cnt = float(N * H * W)
             if process_group is not None and dist.is_initialized() \
                     and dist.get_world_size(process_group) > 1:
-                packed = torch.cat([s, sq, count])
+                packed = torch.cat([s, sq])
                 dist.all_reduce(packed, op=dist.ReduceOp.SUM, group=process_group)
-                s, sq, count = packed[:C], packed[C:2 * C], packed[2 * C:]
-            cnt = count.item()
+                s, sq = packed[:C], packed[C:2 * C]
+                cnt *= dist.get_world_size(process_group)
             mean = s / cnt
             var = sq / cnt - mean * mean
             var = var.clamp_min_(0.0)

@@ -35,7 +35,7 @@ def test_loss_decreases_on_fixed_batch():
         loss.backward()
         model.finalize_backward()
         opt.step()
-        losses.append(float(loss))
+        losses.append(float(loss.detach()))
     assert losses[-1] < losses[0] * 0.7, losses
 
 

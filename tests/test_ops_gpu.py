@@ -87,10 +87,13 @@ def test_fused_bn_relu_autograd_vs_torch(training, dtype):
     y.backward(g.to(dtype))
     y_ref.backward(g)
     assert torch.allclose(x1.grad.float(), x2.grad, **tol(dtype))
-    assert torch.allclose(bn.weight.grad, ref_bn.weight.grad,
-                          atol=1e-2, rtol=1e-2)
-    assert torch.allclose(bn.bias.grad, ref_bn.bias.grad,
-                          atol=1e-2, rtol=1e-2)
+    # dgamma/dbeta are O(sqrt(N*S))-term sums of bf16-quantised products while
+    # the reference sums fp32 copies of the same inputs — allow absolute noise
+    # proportional to the dtype's input quantisation, not a kernel tolerance.
+    gtol = dict(atol=1e-3, rtol=1e-3) if dtype == torch.float32 else \
+        dict(atol=2e-1, rtol=2e-2)
+    assert torch.allclose(bn.weight.grad, ref_bn.weight.grad, **gtol)
+    assert torch.allclose(bn.bias.grad, ref_bn.bias.grad, **gtol)
     if training:
         assert torch.allclose(bn.running_mean, ref_bn.running_mean, atol=1e-3)
         assert torch.allclose(bn.running_var, ref_bn.running_var, atol=1e-3)

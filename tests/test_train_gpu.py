@@ -73,8 +73,67 @@ def test_native_vs_eager_loss_parity():
 
     native = run(False)
     eager = run(True)
-    for a, b in zip(native, eager):
-        assert abs(a - b) < 0.05, (native, eager)
+    # fp32-roundoff-level per-step differences compound through the training
+    # dynamics; allow drift that grows with step count.
+    for i, (a, b) in enumerate(zip(native, eager)):
+        assert abs(a - b) < 0.01 + 0.025 * i, (i, native, eager)
+
+
+def test_hip_graph_step_matches_eager():
+    """3 steps through the captured graph == 3 eager steps (same data/seed)."""
+    from mi355x_ddp.config import TrainConfig
+    from mi355x_ddp.core.amp import autocast_ctx
+    from mi355x_ddp.core.graphs import GraphedTrainStep
+    from mi355x_ddp.core.worker import build_training, init_seeds
+
+    device = torch.device("cuda", 0)
+    gen = torch.Generator().manual_seed(3)
+    data = [(torch.randn(16, 3, 32, 32, generator=gen).pin_memory(),
+             torch.randint(0, 100, (16,), generator=gen).pin_memory())
+            for _ in range(3)]
+
+    def eager():
+        init_seeds(0)
+        cfg = TrainConfig(batch_size=16, amp="fp32", sync_bn=False)
+        model, crit, opt, _, _ = build_training(cfg, device, 1, 0, wrap="flat")
+        model.train()
+        for img, lbl in data:
+            model.zero_grad_buffer()
+            with autocast_ctx("fp32", "cuda"):
+                loss = crit(model(img.to(device)), lbl.to(device))
+            loss.backward()
+            model.finalize_backward()
+            opt.step()
+        return [p.detach().clone() for p in model.parameters()]
+
+    def graphed():
+        init_seeds(0)
+        cfg = TrainConfig(batch_size=16, amp="fp32", sync_bn=False,
+                          hip_graph=True)
+        model, crit, opt, _, _ = build_training(cfg, device, 1, 0, wrap="flat")
+        model.train()
+        # snapshot params before GraphedTrainStep's warmup mutates them,
+        # then restore so both runs start identically
+        snap = [p.detach().clone() for p in model.parameters()]
+        bufs = [b.detach().clone() for b in model.buffers()]
+        step = GraphedTrainStep(model, crit, opt, cfg, device, batch=16)
+        with torch.no_grad():
+            for p, s in zip(model.parameters(), snap):
+                p.copy_(s)
+            for b, s in zip(model.buffers(), bufs):
+                b.copy_(s)
+            for st in opt.state.values():
+                if "momentum_buffer" in st:
+                    st["momentum_buffer"].zero_()
+        for img, lbl in data:
+            step.run(img, lbl)
+        torch.cuda.synchronize()
+        return [p.detach().clone() for p in model.parameters()]
+
+    pe = eager()
+    pg = graphed()
+    for a, b in zip(pe, pg):
+        assert torch.allclose(a, b, atol=2e-4, rtol=1e-4), (a - b).abs().max()
 
 
 def test_bench_single_gpu_contract():
