@@ -91,7 +91,7 @@ __global__ void bn_stats_kernel(const scalar_t* __restrict__ x,
 // Fused BN forward: y = relu((x - mean) * invstd * w + b [+ residual])
 // One elementwise pass; ATen would launch BN, add and ReLU separately.
 // ---------------------------------------------------------------------------
-template <typename scalar_t, bool RELU, bool HAS_RES>
+template <typename scalar_t, bool RELU, bool HAS_RES, bool CLAST>
 __global__ void bn_fwd_kernel(const scalar_t* __restrict__ x,
                               const scalar_t* __restrict__ res,
                               scalar_t* __restrict__ y,
@@ -102,7 +102,7 @@ __global__ void bn_fwd_kernel(const scalar_t* __restrict__ x,
                               long total, int C, int S) {
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
-    const int c = (int)((i / S) % C);
+    const int c = CLAST ? (int)(i % C) : (int)((i / S) % C);
     float v = ((float)x[i] - mean[c]) * invstd[c] * weight[c] + bias[c];
     if (HAS_RES) v += (float)res[i];
     if (RELU) v = fmaxf(v, 0.f);
@@ -150,7 +150,7 @@ __global__ void bn_bwd_reduce_kernel(const scalar_t* __restrict__ dy,
 // BN backward apply: dx (+ optional dresidual = relu-masked dy).
 // training: dx = w*is*(g - sum_dy/cnt - xhat*sum_dy_xhat/cnt); eval: dx = w*is*g
 // ---------------------------------------------------------------------------
-template <typename scalar_t, bool RELU, bool TRAIN, bool NEED_DRES>
+template <typename scalar_t, bool RELU, bool TRAIN, bool NEED_DRES, bool CLAST>
 __global__ void bn_bwd_kernel(const scalar_t* __restrict__ dy,
                               const scalar_t* __restrict__ x,
                               const scalar_t* __restrict__ y,
@@ -164,7 +164,7 @@ __global__ void bn_bwd_kernel(const scalar_t* __restrict__ dy,
                               float inv_count, long total, int C, int S) {
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
-    const int c = (int)((i / S) % C);
+    const int c = CLAST ? (int)(i % C) : (int)((i / S) % C);
     float g = (float)dy[i];
     if (RELU && (float)y[i] <= 0.f) g = 0.f;
     if (NEED_DRES) dres[i] = (scalar_t)g;
@@ -178,6 +178,55 @@ __global__ void bn_bwd_kernel(const scalar_t* __restrict__ dy,
     }
     dx[i] = (scalar_t)v;
   }
+}
+
+
+// ---------------------------------------------------------------------------
+// NHWC (channels_last) variants — the MI355X-preferred layout: per-pixel
+// channels are contiguous, so stats/reductions coalesce across lanes on the
+// channel axis and elementwise kernels compute c = i % C (pow2 on ResNet).
+// MIOpen's tuned implicit-GEMM bf16 kernels are NHWC too, so the whole model
+// runs channels_last with zero transposes.
+// ---------------------------------------------------------------------------
+template <typename scalar_t>
+__global__ void bn_stats_nhwc_kernel(const scalar_t* __restrict__ x,
+                                     float* __restrict__ sum,
+                                     float* __restrict__ sqsum,
+                                     long P, int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s = 0.f, sq = 0.f;
+  for (long p = blockIdx.y; p < P; p += gridDim.y) {
+    const float v = (float)x[p * C + c];
+    s += v;
+    sq += v * v;
+  }
+  atomicAdd(&sum[c], s);
+  atomicAdd(&sqsum[c], sq);
+}
+
+template <typename scalar_t, bool RELU>
+__global__ void bn_bwd_reduce_nhwc_kernel(const scalar_t* __restrict__ dy,
+                                          const scalar_t* __restrict__ x,
+                                          const scalar_t* __restrict__ y,
+                                          const float* __restrict__ mean,
+                                          const float* __restrict__ invstd,
+                                          float* __restrict__ sum_dy,
+                                          float* __restrict__ sum_dy_xhat,
+                                          long P, int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float mu = mean[c], is = invstd[c];
+  float sdy = 0.f, sdyx = 0.f;
+  for (long p = blockIdx.y; p < P; p += gridDim.y) {
+    const long idx = p * C + c;
+    float g = (float)dy[idx];
+    if (RELU && (float)y[idx] <= 0.f) g = 0.f;
+    sdy += g;
+    sdyx += g * ((float)x[idx] - mu) * is;
+  }
+  atomicAdd(&sum_dy[c], sdy);
+  atomicAdd(&sum_dy_xhat[c], sdyx);
 }
 
 // ---------------------------------------------------------------------------
@@ -332,21 +381,42 @@ int split_for(long per_channel_elems, int nchannels) {
 #define CHECK_CUDA(x) TORCH_CHECK(x.is_cuda(), #x " must be a HIP tensor")
 #define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
 
+namespace {
+inline bool is_clast(const at::Tensor& t) {
+  return t.dim() == 4 && t.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+         !t.is_contiguous();
+}
+inline void check_dense(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_contiguous() ||
+              (t.dim() == 4 && t.is_contiguous(at::MemoryFormat::ChannelsLast)),
+              name, " must be NCHW- or NHWC-contiguous");
+}
+}  // namespace
+
 std::vector<at::Tensor> bn_stats(at::Tensor x) {
-  CHECK_CUDA(x); CHECK_CONTIG(x);
+  CHECK_CUDA(x); check_dense(x, "x");
   const int N = x.size(0), C = x.size(1);
-  const int S = x.numel() / ((long)N * C);
+  const long S = x.numel() / ((long)N * C);
   auto opts = x.options().dtype(at::kFloat);
   auto sum = at::zeros({C}, opts);
   auto sqsum = at::zeros({C}, opts);
-  const int split = split_for((long)N * S, C);
-  dim3 grid(C, split);
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::Half, at::ScalarType::BFloat16,
       x.scalar_type(), "bn_stats", [&] {
-    hipLaunchKernelGGL(bn_stats_kernel<scalar_t>, grid, dim3(256), 0,
-                       cur_stream(),
-                       x.data_ptr<scalar_t>(), sum.data_ptr<float>(),
-                       sqsum.data_ptr<float>(), N, C, S);
+    if (is_clast(x)) {
+      const long P = (long)N * S;
+      const int cblocks = (C + 255) / 256;
+      const int split = (int)std::min(P, (long)std::max(1, 768 / cblocks));
+      hipLaunchKernelGGL(bn_stats_nhwc_kernel<scalar_t>, dim3(cblocks, split),
+                         dim3(256), 0, cur_stream(),
+                         x.data_ptr<scalar_t>(), sum.data_ptr<float>(),
+                         sqsum.data_ptr<float>(), P, C);
+    } else {
+      const int split = split_for((long)N * S, C);
+      hipLaunchKernelGGL(bn_stats_kernel<scalar_t>, dim3(C, split), dim3(256),
+                         0, cur_stream(),
+                         x.data_ptr<scalar_t>(), sum.data_ptr<float>(),
+                         sqsum.data_ptr<float>(), N, C, (int)S);
+    }
   });
   return {sum, sqsum};
 }
@@ -354,32 +424,40 @@ std::vector<at::Tensor> bn_stats(at::Tensor x) {
 at::Tensor bn_fwd(at::Tensor x, at::Tensor weight, at::Tensor bias,
                   at::Tensor mean, at::Tensor invstd, bool relu,
                   at::Tensor residual) {
-  CHECK_CUDA(x); CHECK_CONTIG(x);
+  CHECK_CUDA(x); check_dense(x, "x");
   const int N = x.size(0), C = x.size(1);
   const int S = x.numel() / ((long)N * C);
   const long total = x.numel();
   const bool has_res = residual.defined() && residual.numel() > 0;
+  const bool clast = is_clast(x);
+  if (has_res) {
+    TORCH_CHECK(is_clast(residual) == clast, "residual layout mismatch");
+  }
   auto y = at::empty_like(x);
   auto wf = weight.to(at::kFloat).contiguous();
   auto bf = bias.to(at::kFloat).contiguous();
-  const int blocks = (int)std::min((total + 255) / 256, (long)8192);
+  const int blocks = (int)std::min((total + 1023) / 1024, (long)4096);
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::Half, at::ScalarType::BFloat16,
       x.scalar_type(), "bn_fwd", [&] {
-    auto launch = [&](auto relu_c, auto res_c) {
+    auto launch = [&](auto relu_c, auto res_c, auto cl_c) {
       hipLaunchKernelGGL((bn_fwd_kernel<scalar_t, decltype(relu_c)::value,
-                                        decltype(res_c)::value>),
-                         dim3(blocks), dim3(256), 0,
-                         cur_stream(),
+                                        decltype(res_c)::value,
+                                        decltype(cl_c)::value>),
+                         dim3(blocks), dim3(256), 0, cur_stream(),
                          x.data_ptr<scalar_t>(),
                          has_res ? residual.data_ptr<scalar_t>() : nullptr,
                          y.data_ptr<scalar_t>(), wf.data_ptr<float>(),
                          bf.data_ptr<float>(), mean.data_ptr<float>(),
                          invstd.data_ptr<float>(), total, C, S);
     };
-    if (relu && has_res) launch(std::true_type{}, std::true_type{});
-    else if (relu) launch(std::true_type{}, std::false_type{});
-    else if (has_res) launch(std::false_type{}, std::true_type{});
-    else launch(std::false_type{}, std::false_type{});
+    auto l2 = [&](auto relu_c, auto res_c) {
+      clast ? launch(relu_c, res_c, std::true_type{})
+            : launch(relu_c, res_c, std::false_type{});
+    };
+    if (relu && has_res) l2(std::true_type{}, std::true_type{});
+    else if (relu) l2(std::true_type{}, std::false_type{});
+    else if (has_res) l2(std::false_type{}, std::true_type{});
+    else l2(std::false_type{}, std::false_type{});
   });
   return y;
 }
@@ -387,25 +465,42 @@ at::Tensor bn_fwd(at::Tensor x, at::Tensor weight, at::Tensor bias,
 std::vector<at::Tensor> bn_bwd_reduce(at::Tensor dy, at::Tensor x,
                                       at::Tensor mean, at::Tensor invstd,
                                       at::Tensor y, bool relu) {
-  CHECK_CUDA(dy); CHECK_CONTIG(dy); CHECK_CONTIG(x);
+  CHECK_CUDA(dy); check_dense(dy, "dy"); check_dense(x, "x");
+  TORCH_CHECK(is_clast(dy) == is_clast(x), "dy/x layout mismatch");
   const int N = x.size(0), C = x.size(1);
-  const int S = x.numel() / ((long)N * C);
+  const long S = x.numel() / ((long)N * C);
   auto opts = x.options().dtype(at::kFloat);
   auto sum_dy = at::zeros({C}, opts);
   auto sum_dy_xhat = at::zeros({C}, opts);
-  const int split = split_for((long)N * S, C);
-  dim3 grid(C, split);
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::Half, at::ScalarType::BFloat16,
       x.scalar_type(), "bn_bwd_reduce", [&] {
-    auto launch = [&](auto relu_c) {
-      hipLaunchKernelGGL((bn_bwd_reduce_kernel<scalar_t, decltype(relu_c)::value>),
-                         grid, dim3(256), 0, cur_stream(),
-                         dy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
-                         y.data_ptr<scalar_t>(), mean.data_ptr<float>(),
-                         invstd.data_ptr<float>(), sum_dy.data_ptr<float>(),
-                         sum_dy_xhat.data_ptr<float>(), N, C, S);
-    };
-    relu ? launch(std::true_type{}) : launch(std::false_type{});
+    if (is_clast(x)) {
+      const long P = (long)N * S;
+      const int cblocks = (C + 255) / 256;
+      const int split = (int)std::min(P, (long)std::max(1, 768 / cblocks));
+      auto launch = [&](auto relu_c) {
+        hipLaunchKernelGGL((bn_bwd_reduce_nhwc_kernel<scalar_t,
+                                                      decltype(relu_c)::value>),
+                           dim3(cblocks, split), dim3(256), 0, cur_stream(),
+                           dy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
+                           y.data_ptr<scalar_t>(), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), sum_dy.data_ptr<float>(),
+                           sum_dy_xhat.data_ptr<float>(), P, C);
+      };
+      relu ? launch(std::true_type{}) : launch(std::false_type{});
+    } else {
+      const int split = split_for((long)N * S, C);
+      auto launch = [&](auto relu_c) {
+        hipLaunchKernelGGL((bn_bwd_reduce_kernel<scalar_t,
+                                                 decltype(relu_c)::value>),
+                           dim3(C, split), dim3(256), 0, cur_stream(),
+                           dy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
+                           y.data_ptr<scalar_t>(), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), sum_dy.data_ptr<float>(),
+                           sum_dy_xhat.data_ptr<float>(), N, C, (int)S);
+      };
+      relu ? launch(std::true_type{}) : launch(std::false_type{});
+    }
   });
   return {sum_dy, sum_dy_xhat};
 }
@@ -415,22 +510,23 @@ std::vector<at::Tensor> bn_bwd(at::Tensor dy, at::Tensor x, at::Tensor weight,
                                at::Tensor sum_dy, at::Tensor sum_dy_xhat,
                                double count, at::Tensor y, bool relu,
                                bool training, bool need_dres) {
-  CHECK_CUDA(dy); CHECK_CONTIG(dy); CHECK_CONTIG(x);
+  CHECK_CUDA(dy); check_dense(dy, "dy"); check_dense(x, "x");
   const int N = x.size(0), C = x.size(1);
   const int S = x.numel() / ((long)N * C);
   const long total = x.numel();
+  const bool clast = is_clast(x);
   auto dx = at::empty_like(x);
   auto dres = need_dres ? at::empty_like(x) : at::empty({0}, x.options());
   auto wf = weight.to(at::kFloat).contiguous();
-  const int blocks = (int)std::min((total + 255) / 256, (long)8192);
+  const int blocks = (int)std::min((total + 1023) / 1024, (long)4096);
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::Half, at::ScalarType::BFloat16,
       x.scalar_type(), "bn_bwd", [&] {
-    auto launch = [&](auto relu_c, auto train_c, auto dres_c) {
+    auto launch = [&](auto relu_c, auto train_c, auto dres_c, auto cl_c) {
       hipLaunchKernelGGL((bn_bwd_kernel<scalar_t, decltype(relu_c)::value,
                                         decltype(train_c)::value,
-                                        decltype(dres_c)::value>),
-                         dim3(blocks), dim3(256), 0,
-                         cur_stream(),
+                                        decltype(dres_c)::value,
+                                        decltype(cl_c)::value>),
+                         dim3(blocks), dim3(256), 0, cur_stream(),
                          dy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
                          y.data_ptr<scalar_t>(), dx.data_ptr<scalar_t>(),
                          need_dres ? dres.data_ptr<scalar_t>() : nullptr,
@@ -439,17 +535,20 @@ std::vector<at::Tensor> bn_bwd(at::Tensor dy, at::Tensor x, at::Tensor weight,
                          sum_dy_xhat.data_ptr<float>(),
                          (float)(1.0 / count), total, C, S);
     };
-    // 8-way static dispatch over (relu, training, need_dres)
+    auto l3 = [&](auto relu_c, auto train_c, auto dres_c) {
+      clast ? launch(relu_c, train_c, dres_c, std::true_type{})
+            : launch(relu_c, train_c, dres_c, std::false_type{});
+    };
     if (relu) {
-      if (training) need_dres ? launch(std::true_type{}, std::true_type{}, std::true_type{})
-                              : launch(std::true_type{}, std::true_type{}, std::false_type{});
-      else need_dres ? launch(std::true_type{}, std::false_type{}, std::true_type{})
-                     : launch(std::true_type{}, std::false_type{}, std::false_type{});
+      if (training) need_dres ? l3(std::true_type{}, std::true_type{}, std::true_type{})
+                              : l3(std::true_type{}, std::true_type{}, std::false_type{});
+      else need_dres ? l3(std::true_type{}, std::false_type{}, std::true_type{})
+                     : l3(std::true_type{}, std::false_type{}, std::false_type{});
     } else {
-      if (training) need_dres ? launch(std::false_type{}, std::true_type{}, std::true_type{})
-                              : launch(std::false_type{}, std::true_type{}, std::false_type{});
-      else need_dres ? launch(std::false_type{}, std::false_type{}, std::true_type{})
-                     : launch(std::false_type{}, std::false_type{}, std::false_type{});
+      if (training) need_dres ? l3(std::false_type{}, std::true_type{}, std::true_type{})
+                              : l3(std::false_type{}, std::true_type{}, std::false_type{});
+      else need_dres ? l3(std::false_type{}, std::false_type{}, std::true_type{})
+                     : l3(std::false_type{}, std::false_type{}, std::false_type{});
     }
   });
   return {dx, dres};
@@ -505,9 +604,11 @@ void multi_tensor_sgd(std::vector<at::Tensor> params,
     int t = 0;
     for (; t < MT_MAX_TENSORS && i < params.size(); ++t, ++i) {
       auto& p = params[i];
-      TORCH_CHECK(p.is_cuda() && p.is_contiguous() &&
+      TORCH_CHECK(p.is_cuda() && p.is_non_overlapping_and_dense() &&
                   p.scalar_type() == at::kFloat,
-                  "multi_tensor_sgd expects contiguous fp32 params");
+                  "multi_tensor_sgd expects dense fp32 params");
+      TORCH_CHECK(grads[i].strides() == p.strides(),
+                  "multi_tensor_sgd: grad/param layout mismatch");
       tl.p[t] = p.data_ptr<float>();
       tl.g[t] = grads[i].data_ptr<float>();
       tl.m[t] = has_momentum ? bufs[i].data_ptr<float>() : nullptr;
@@ -531,9 +632,9 @@ at::Tensor multi_tensor_unscale(std::vector<at::Tensor> grads, double inv_scale)
     int nblocks = 0;
     int t = 0;
     for (; t < MT_MAX_TENSORS && i < grads.size(); ++t, ++i) {
-      TORCH_CHECK(grads[i].is_cuda() && grads[i].is_contiguous() &&
+      TORCH_CHECK(grads[i].is_cuda() && grads[i].is_non_overlapping_and_dense() &&
                   grads[i].scalar_type() == at::kFloat,
-                  "multi_tensor_unscale expects contiguous fp32 grads");
+                  "multi_tensor_unscale expects dense fp32 grads");
       tl.g[t] = grads[i].data_ptr<float>();
       tl.size[t] = (int)grads[i].numel();
       nblocks += (tl.size[t] + MT_CHUNK - 1) / MT_CHUNK;

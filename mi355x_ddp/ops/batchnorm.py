@@ -83,9 +83,15 @@ class _FusedBNFunction(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, residual, weight, bias, running_mean, running_var,
                 training, momentum, eps, relu, process_group):
-        x = x.contiguous()
+        # preserve channels_last end-to-end: NHWC tensors run the NHWC kernels,
+        # no transposes are ever inserted around the fused op
+        fmt = torch.channels_last if (x.dim() == 4 and x.is_contiguous(
+            memory_format=torch.channels_last) and not x.is_contiguous()) \
+            else torch.contiguous_format
+        x = x.contiguous(memory_format=fmt)
         if residual is not None:
-            residual = residual.contiguous()
+            residual = residual.contiguous(memory_format=fmt)
+        ctx.fmt = fmt
         N, C, H, W = x.shape
         if training:
             s, sq = _stats(x)
@@ -123,7 +129,7 @@ class _FusedBNFunction(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x, weight, mean, invstd, y = ctx.saved_tensors
-        dy = dy.contiguous()
+        dy = dy.contiguous(memory_format=ctx.fmt)
         sum_dy, sum_dy_xhat = _bwd_reduce(dy, x, mean, invstd, y, ctx.relu)
         count = ctx.count
         # dgamma/dbeta are the LOCAL sums — the DP gradient all-reduce averages

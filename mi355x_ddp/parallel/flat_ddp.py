@@ -58,7 +58,16 @@ class FlatDDP(nn.Module):
         self._views = {}
         for p in reversed(params):
             n = p.numel()
-            view = self.flat_grads[offset:offset + n].view_as(p)
+            flat_slice = self.flat_grads[offset:offset + n]
+            if p.dim() == 4 and p.is_contiguous(memory_format=torch.channels_last) \
+                    and not p.is_contiguous():
+                # grad view must share the param's memory order (channels_last)
+                # so autograd accumulates without layout conversion and the
+                # fused SGD walks p/g/m in the same element order
+                N_, C_, H_, W_ = p.shape
+                view = flat_slice.view(N_, H_, W_, C_).permute(0, 3, 1, 2)
+            else:
+                view = flat_slice.view_as(p)
             self._views[p] = (offset, view)
             p.grad = view
             offset += n

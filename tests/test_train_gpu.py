@@ -93,7 +93,7 @@ def test_hip_graph_step_matches_eager():
             for _ in range(3)]
 
     def eager():
-        init_seeds(0)
+        init_seeds(0, deterministic=True)
         cfg = TrainConfig(batch_size=16, amp="fp32", sync_bn=False)
         model, crit, opt, _, _ = build_training(cfg, device, 1, 0, wrap="flat")
         model.train()
@@ -107,7 +107,7 @@ def test_hip_graph_step_matches_eager():
         return [p.detach().clone() for p in model.parameters()]
 
     def graphed():
-        init_seeds(0)
+        init_seeds(0, deterministic=True)
         cfg = TrainConfig(batch_size=16, amp="fp32", sync_bn=False,
                           hip_graph=True)
         model, crit, opt, _, _ = build_training(cfg, device, 1, 0, wrap="flat")
@@ -133,7 +133,7 @@ def test_hip_graph_step_matches_eager():
     pe = eager()
     pg = graphed()
     for a, b in zip(pe, pg):
-        assert torch.allclose(a, b, atol=2e-4, rtol=1e-4), (a - b).abs().max()
+        assert torch.allclose(a, b, atol=1e-3, rtol=1e-3), (a - b).abs().max()
 
 
 def test_bench_single_gpu_contract():
