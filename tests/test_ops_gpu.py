@@ -122,6 +122,84 @@ def test_fused_bn_add_relu_autograd_vs_torch(dtype):
     assert torch.allclose(r1.grad.float(), r2.grad, **tol(dtype))
 
 
+@pytest.mark.parametrize("shape", SHAPES)
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_bn_stats_kernel_nhwc(shape, dtype):
+    from mi355x_ddp.ops import _backend
+    x = torch.randn(*shape, device=DEV, dtype=dtype) \
+        .to(memory_format=torch.channels_last)
+    s, sq = _backend.C().bn_stats(x)
+    xf = x.float()
+    assert torch.allclose(s, xf.sum((0, 2, 3)), **tol(dtype))
+    assert torch.allclose(sq, (xf * xf).sum((0, 2, 3)), **tol(dtype))
+
+
+@pytest.mark.parametrize("training", [True, False])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_fused_bn_relu_nhwc_vs_torch(training, dtype):
+    """channels_last inputs run the NHWC kernels; reference is fp32 NCHW."""
+    from mi355x_ddp.ops import bn_relu
+    torch.manual_seed(0)
+    bn = nn.BatchNorm2d(64).to(DEV)
+    ref_bn = nn.BatchNorm2d(64).to(DEV)
+    ref_bn.load_state_dict(bn.state_dict())
+    bn.train(training), ref_bn.train(training)
+
+    x1 = torch.randn(8, 64, 16, 16, device=DEV, dtype=dtype) \
+        .to(memory_format=torch.channels_last).requires_grad_(True)
+    x2 = x1.detach().float().contiguous().requires_grad_(True)
+
+    y = bn_relu(x1, bn)
+    assert y.is_contiguous(memory_format=torch.channels_last)
+    y_ref = F.relu(ref_bn(x2))
+    assert torch.allclose(y.float().contiguous(), y_ref, **tol(dtype))
+
+    g = torch.randn_like(y_ref)
+    y.backward(g.to(dtype).to(memory_format=torch.channels_last))
+    y_ref.backward(g)
+    assert torch.allclose(x1.grad.float().contiguous(), x2.grad, **tol(dtype))
+    gtol = dict(atol=1e-3, rtol=1e-3) if dtype == torch.float32 else \
+        dict(atol=2e-1, rtol=2e-2)
+    assert torch.allclose(bn.weight.grad, ref_bn.weight.grad, **gtol)
+    assert torch.allclose(bn.bias.grad, ref_bn.bias.grad, **gtol)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_fused_bn_add_relu_nhwc_vs_torch(dtype):
+    from mi355x_ddp.ops import bn_add_relu
+    torch.manual_seed(1)
+    bn = nn.BatchNorm2d(128).to(DEV)
+    ref_bn = nn.BatchNorm2d(128).to(DEV)
+    ref_bn.load_state_dict(bn.state_dict())
+    x1 = torch.randn(4, 128, 8, 8, device=DEV, dtype=dtype) \
+        .to(memory_format=torch.channels_last).requires_grad_(True)
+    r1 = torch.randn(4, 128, 8, 8, device=DEV, dtype=dtype) \
+        .to(memory_format=torch.channels_last).requires_grad_(True)
+    x2 = x1.detach().float().contiguous().requires_grad_(True)
+    r2 = r1.detach().float().contiguous().requires_grad_(True)
+
+    y = bn_add_relu(x1, r1, bn)
+    y_ref = F.relu(ref_bn(x2) + r2)
+    assert torch.allclose(y.float().contiguous(), y_ref, **tol(dtype))
+    g = torch.randn_like(y_ref)
+    y.backward(g.to(dtype).to(memory_format=torch.channels_last))
+    y_ref.backward(g)
+    assert torch.allclose(x1.grad.float().contiguous(), x2.grad, **tol(dtype))
+    assert torch.allclose(r1.grad.float().contiguous(), r2.grad, **tol(dtype))
+
+
+def test_multi_tensor_sgd_channels_last():
+    from mi355x_ddp.ops import _backend
+    torch.manual_seed(5)
+    p = torch.randn(16, 8, 3, 3, device=DEV).to(memory_format=torch.channels_last)
+    g = torch.randn_like(p)  # preserves channels_last
+    m = torch.zeros_like(p)
+    ref = p.detach().clone()
+    _backend.C().multi_tensor_sgd([p], [g], [m], 0.1, 0.9, 1e-4)
+    d = g.add(ref, alpha=1e-4)
+    assert torch.allclose(p, ref - 0.1 * d, atol=1e-6)
+
+
 @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
 @pytest.mark.parametrize("nc", [(16, 100), (256, 100), (64, 1000)])
 def test_xent_kernel_vs_torch(dtype, nc):
