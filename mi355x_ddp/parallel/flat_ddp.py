@@ -59,15 +59,10 @@ class FlatDDP(nn.Module):
         for p in reversed(params):
             n = p.numel()
             flat_slice = self.flat_grads[offset:offset + n]
-            if p.dim() == 4 and p.is_contiguous(memory_format=torch.channels_last) \
-                    and not p.is_contiguous():
-                # grad view must share the param's memory order (channels_last)
-                # so autograd accumulates without layout conversion and the
-                # fused SGD walks p/g/m in the same element order
-                N_, C_, H_, W_ = p.shape
-                view = flat_slice.view(N_, H_, W_, C_).permute(0, 3, 1, 2)
-            else:
-                view = flat_slice.view_as(p)
+            # the grad view copies the param's EXACT strides (NCHW, NHWC or the
+            # ambiguous 1x1-conv case) so autograd accumulates without layout
+            # conversion and the fused SGD walks p/g/m in one memory order
+            view = flat_slice.as_strided(p.shape, p.stride())
             self._views[p] = (offset, view)
             p.grad = view
             offset += n
