@@ -15,7 +15,8 @@ import sys
 import sysconfig
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-SOURCES = [os.path.join(REPO, "csrc", "mi355x_kernels.hip")]
+SOURCES = [os.path.join(REPO, "csrc", "mi355x_kernels.hip"),
+           os.path.join(REPO, "csrc", "conv_igemm.hip")]
 
 
 def out_path() -> str:

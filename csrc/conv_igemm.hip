@@ -1,0 +1,416 @@
+// MI355X (gfx950/CDNA4) implicit-GEMM convolution kernels — NHWC bf16.
+//
+// Replaces the reference's dependency on cuDNN conv kernels (SURVEY.md §2.2
+// N1: 3x3 s1/s2 and 1x1 convs of the CIFAR ResNet, fwd + dgrad + wgrad).
+// Written MFMA-first for CDNA4: v_mfma_f32_16x16x32_bf16 tiles, fp32
+// accumulation, LDS-staged operand tiles sized for 64-wide wavefronts.
+//
+// GEMM views (all NHWC, reduction in fp32):
+//   fwd  : out[N*P*Q][Kout] = im2col(x)[M][R*S*C]   @ w[Kout][R*S*C]^T
+//   dgrad: dx[N*H*W][C]     = im2col'(dy)[M][R*S*K] @ wT[C][R*S*K]^T
+//          (wT = weight rotated 180° and transposed to [C][R][S][K], built
+//           host-side — a few KB)
+//   wgrad: dw[Kout][R*S*C]  = dy^T[Kout][N*P*Q]     @ im2col(x)[N*P*Q][R*S*C]
+//          (split-K over N*P*Q chunks, fp32 atomics into a dw accumulator)
+//
+// FAST path: when the contiguous channel count is a multiple of BK(=64) a
+// BK-sized reduction chunk lies inside ONE (r,s) filter tap with a contiguous
+// channel run — every A-tile row is one 64-byte-aligned 128 B global read
+// (or zero-fill). This covers every ResNet conv except the 3-channel stem,
+// which takes the generic per-element gather path.
+#include <torch/extension.h>
+#include <ATen/ATen.h>
+#include <hip/hip_runtime.h>
+#include <ATen/hip/HIPContext.h>
+#include <c10/hip/HIPStream.h>
+
+#include <vector>
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+constexpr int BM = 128;   // GEMM M tile (output pixels)
+constexpr int BN = 64;    // GEMM N tile (output channels / rsc)
+constexpr int BK = 64;    // reduction tile
+constexpr int LDK = BK + 8;  // padded LDS row stride: (m*LDK) covers all 64
+                             // banks across a 16-lane ds_read_b128 group
+
+struct ConvDims {
+  int Nb;                // batch
+  int OH, OW;            // spatial dims of the GEMM-M tensor (out / in)
+  int GH, GW, GC;        // dims of the gathered tensor (x for fwd, dy for dgrad)
+  int R, S;              // filter
+  int stride, pad;
+  int M, N, K;           // GEMM sizes: K = R*S*GC
+};
+
+// Decode a flat GEMM-M row into (image, row, col).
+__device__ __forceinline__ void decode_m(int m, const ConvDims& d,
+                                         int& n, int& oh, int& ow) {
+  ow = m % d.OW;
+  const int t = m / d.OW;
+  oh = t % d.OH;
+  n = t / d.OH;
+}
+
+// Input row coordinate for filter tap i. MODE 0 = fwd (oh is an output pixel,
+// gather from x), MODE 1 = dgrad (oh is an input pixel, gather from dy with
+// the transpose-conv index relation; tap index i is the 180°-rotated r).
+template <int MODE>
+__device__ __forceinline__ bool tap_coord(int oh, int i, int filt, int stride,
+                                          int pad, int lim, int& ih) {
+  if (MODE == 0) {
+    ih = oh * stride + i - pad;
+    return ih >= 0 && ih < lim;
+  }
+  const int t = oh + pad - filt + 1 + i;
+  if (t < 0 || t % stride != 0) return false;
+  ih = t / stride;
+  return ih < lim;
+}
+
+// ---------------------------------------------------------------------------
+// fwd / dgrad kernel: out[M][N] = gatherA[M][K] @ B[N][K]^T
+// 256 threads = 4 waves as a 2x2 wave grid; per wave 64x32 via 4x2 MFMA tiles.
+// ---------------------------------------------------------------------------
+template <int MODE, bool FAST>
+__global__ __launch_bounds__(256)
+void conv_igemm_kernel(const __bf16* __restrict__ Ag,
+                       const __bf16* __restrict__ Bg,
+                       __bf16* __restrict__ out, ConvDims d) {
+  __shared__ __bf16 sA[BM * LDK];
+  __shared__ __bf16 sB[BN * LDK];
+
+  const int tid = threadIdx.x;
+  const int m0 = blockIdx.y * BM;
+  const int n0 = blockIdx.x * BN;
+
+  // A loader: thread -> (row, 32-element half of the BK chunk)
+  const int a_row = tid >> 1;
+  const int a_off = (tid & 1) * 32;
+  int a_n, a_oh, a_ow;
+  {
+    int m = m0 + a_row;
+    if (m >= d.M) m = d.M - 1;  // clamped rows only feed predicated-out outputs
+    decode_m(m, d, a_n, a_oh, a_ow);
+  }
+  // B loader: thread -> (row, 16-element quarter of the BK chunk)
+  const int b_row = tid & 63;
+  const int b_off = (tid >> 6) * 16;
+  const long b_base = (long)min(n0 + b_row, d.N - 1) * d.K;
+
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wm = (wave >> 1) * 64;
+  const int wn = (wave & 1) * 32;
+  const int fr = lane & 15;
+  const int fk = (lane >> 4) * 8;
+
+  f32x4 acc[4][2] = {};
+
+  const int SC = d.S * d.GC;
+  for (int kk0 = 0; kk0 < d.K; kk0 += BK) {
+    // ---- stage A tile -------------------------------------------------
+    if (FAST) {
+      const int i = kk0 / SC;
+      const int rem = kk0 - i * SC;
+      const int j = rem / d.GC;
+      const int c0 = rem - j * d.GC + a_off;
+      int ih, iw;
+      const bool ok = tap_coord<MODE>(a_oh, i, d.R, d.stride, d.pad, d.GH, ih)
+                    & tap_coord<MODE>(a_ow, j, d.S, d.stride, d.pad, d.GW, iw);
+      float4 v0 = {}, v1 = {}, v2 = {}, v3 = {};
+      if (ok) {
+        const float4* src = (const float4*)(Ag +
+            (((long)a_n * d.GH + ih) * d.GW + iw) * d.GC + c0);
+        v0 = src[0]; v1 = src[1]; v2 = src[2]; v3 = src[3];
+      }
+      float4* dst = (float4*)(sA + a_row * LDK + a_off);
+      dst[0] = v0; dst[1] = v1; dst[2] = v2; dst[3] = v3;
+    } else {
+      // generic gather: one element at a time (stem conv only)
+      for (int e = tid; e < BM * BK; e += 256) {
+        const int row = e >> 6, kk = kk0 + (e & 63);
+        __bf16 v = (__bf16)0.f;
+        if (kk < d.K) {
+          int m = m0 + row;
+          if (m >= d.M) m = d.M - 1;
+          int n, oh, ow;
+          decode_m(m, d, n, oh, ow);
+          const int i = kk / SC, rem = kk - i * SC;
+          const int j = rem / d.GC, c = rem - j * d.GC;
+          int ih, iw;
+          if (tap_coord<MODE>(oh, i, d.R, d.stride, d.pad, d.GH, ih) &&
+              tap_coord<MODE>(ow, j, d.S, d.stride, d.pad, d.GW, iw))
+            v = Ag[(((long)n * d.GH + ih) * d.GW + iw) * d.GC + c];
+        }
+        sA[row * LDK + (e & 63)] = v;
+      }
+    }
+    // ---- stage B tile (dense rows of w / wT) ---------------------------
+    if (FAST) {
+      const float4* src = (const float4*)(Bg + b_base + kk0 + b_off);
+      float4* dst = (float4*)(sB + b_row * LDK + b_off);
+      dst[0] = src[0]; dst[1] = src[1];
+    } else {
+      for (int e = tid; e < BN * BK; e += 256) {
+        const int row = e >> 6, kk = kk0 + (e & 63);
+        sB[row * LDK + (e & 63)] = (kk < d.K)
+            ? Bg[(long)min(n0 + row, d.N - 1) * d.K + kk] : (__bf16)0.f;
+      }
+    }
+    __syncthreads();
+
+    #pragma unroll
+    for (int ks = 0; ks < BK; ks += 32) {
+      bf16x8 af[4], bf[2];
+      #pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        af[mi] = *(const bf16x8*)&sA[(wm + mi * 16 + fr) * LDK + ks + fk];
+      #pragma unroll
+      for (int ni = 0; ni < 2; ++ni)
+        bf[ni] = *(const bf16x8*)&sB[(wn + ni * 16 + fr) * LDK + ks + fk];
+      #pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        #pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: D row = (lane>>4)*4 + reg, col = lane&15 ----------------
+  const int dm = (lane >> 4) * 4;
+  const int dn = lane & 15;
+  #pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+    #pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      const int n = n0 + wn + ni * 16 + dn;
+      if (n >= d.N) continue;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = m0 + wm + mi * 16 + dm + r;
+        if (m < d.M) out[(long)m * d.N + n] = (__bf16)acc[mi][ni][r];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// wgrad kernel: dw[Kout][RSC] += dy^T[Kout][NPQ-chunk] @ im2col(x)[chunk][RSC]
+// 64x64 output tile per block, 2x2 waves of 32x32; split-K over NPQ chunks
+// with fp32 atomics. Both LDS tiles are written transposed (reduction index
+// contiguous per row) so MFMA fragment reads are 16-byte ds_read_b128.
+// ---------------------------------------------------------------------------
+template <bool FAST>
+__global__ __launch_bounds__(256)
+void conv_wgrad_kernel(const __bf16* __restrict__ dy,
+                       const __bf16* __restrict__ x,
+                       float* __restrict__ dw, ConvDims d) {
+  // d: M = Kout, N = R*S*C, K = Nb*P*Q; OH/OW = P,Q; GH/GW/GC = H,W,C
+  __shared__ __bf16 sA[64 * LDK];  // [kout][npq]
+  __shared__ __bf16 sB[64 * LDK];  // [rsc][npq]
+
+  const int tid = threadIdx.x;
+  const int m0 = blockIdx.y * 64;
+  const int n0 = blockIdx.x * 64;
+  const int row = tid & 63;       // npq row within the chunk
+  const int grp = (tid >> 6) * 16;  // 16-element column group
+
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wm = (wave >> 1) * 32;
+  const int wn = (wave & 1) * 32;
+  const int fr = lane & 15;
+  const int fk = (lane >> 4) * 8;
+
+  const int SC = d.S * d.GC;
+  const int nchunks = (d.K + BK - 1) / BK;
+  f32x4 acc[2][2] = {};
+
+  for (int kc = blockIdx.z; kc < nchunks; kc += gridDim.z) {
+    const int kk0 = kc * BK;
+    const int npq = kk0 + row;
+    int xn, xp, xq;
+    decode_m(min(npq, d.K - 1), d, xn, xp, xq);
+    const bool npq_ok = npq < d.K;
+
+    // ---- dy tile, transposed into sA[kout][npq] ------------------------
+    {
+      __bf16 vals[16];
+      if (npq_ok && (FAST || m0 + grp + 16 <= d.M)) {
+        const float4* src = (const float4*)(dy + (long)npq * d.M + m0 + grp);
+        *(float4*)&vals[0] = src[0];
+        *(float4*)&vals[8] = src[1];
+      } else {
+        #pragma unroll
+        for (int e = 0; e < 16; ++e)
+          vals[e] = (npq_ok && m0 + grp + e < d.M)
+              ? dy[(long)npq * d.M + m0 + grp + e] : (__bf16)0.f;
+      }
+      #pragma unroll
+      for (int e = 0; e < 16; ++e)
+        sA[(grp + e) * LDK + row] = vals[e];
+    }
+    // ---- x tile, transposed into sB[rsc][npq] --------------------------
+    {
+      __bf16 vals[16];
+      if (FAST) {
+        const int i = n0 / SC;
+        const int rem = n0 - i * SC;
+        const int j = rem / d.GC;
+        const int c0 = rem - j * d.GC + grp;
+        const int ih = xp * d.stride + i - d.pad;
+        const int iw = xq * d.stride + j - d.pad;
+        if (npq_ok && ih >= 0 && ih < d.GH && iw >= 0 && iw < d.GW) {
+          const float4* src = (const float4*)(x +
+              (((long)xn * d.GH + ih) * d.GW + iw) * d.GC + c0);
+          *(float4*)&vals[0] = src[0];
+          *(float4*)&vals[8] = src[1];
+        } else {
+          #pragma unroll
+          for (int e = 0; e < 16; ++e) vals[e] = (__bf16)0.f;
+        }
+      } else {
+        #pragma unroll
+        for (int e = 0; e < 16; ++e) {
+          const int nn = n0 + grp + e;
+          __bf16 v = (__bf16)0.f;
+          if (npq_ok && nn < d.N) {
+            const int i = nn / SC, rem = nn - i * SC;
+            const int j = rem / d.GC, c = rem - j * d.GC;
+            const int ih = xp * d.stride + i - d.pad;
+            const int iw = xq * d.stride + j - d.pad;
+            if (ih >= 0 && ih < d.GH && iw >= 0 && iw < d.GW)
+              v = x[(((long)xn * d.GH + ih) * d.GW + iw) * d.GC + c];
+          }
+          vals[e] = v;
+        }
+      }
+      #pragma unroll
+      for (int e = 0; e < 16; ++e)
+        sB[(grp + e) * LDK + row] = vals[e];
+    }
+    __syncthreads();
+
+    #pragma unroll
+    for (int ks = 0; ks < BK; ks += 32) {
+      bf16x8 af[2], bf[2];
+      #pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+        af[mi] = *(const bf16x8*)&sA[(wm + mi * 16 + fr) * LDK + ks + fk];
+      #pragma unroll
+      for (int ni = 0; ni < 2; ++ni)
+        bf[ni] = *(const bf16x8*)&sB[(wn + ni * 16 + fr) * LDK + ks + fk];
+      #pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+        #pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int dm = (lane >> 4) * 4;
+  const int dn = lane & 15;
+  #pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+    #pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      const int n = n0 + wn + ni * 16 + dn;
+      if (n >= d.N) continue;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = m0 + wm + mi * 16 + dm + r;
+        if (m < d.M) atomicAdd(&dw[(long)m * d.N + n], acc[mi][ni][r]);
+      }
+    }
+}
+
+hipStream_t conv_stream() { return c10::hip::getCurrentHIPStream().stream(); }
+
+inline void check_nhwc_bf16(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == at::ScalarType::BFloat16,
+              name, " must be a bf16 HIP tensor");
+  TORCH_CHECK(t.dim() == 4 &&
+              (t.is_contiguous(at::MemoryFormat::ChannelsLast) ||
+               t.is_contiguous()),
+              name, " must be 4-D and dense");
+}
+
+inline const __bf16* bf16_ptr(const at::Tensor& t) {
+  return reinterpret_cast<const __bf16*>(t.data_ptr());
+}
+
+}  // namespace
+
+// out (N,K,P,Q) channels_last <- x (N,C,H,W) channels_last, w (K,C,R,S)
+// channels_last (memory [K][R][S][C]).
+at::Tensor conv_fwd_igemm(at::Tensor x, at::Tensor w, long stride, long pad) {
+  check_nhwc_bf16(x, "x"); check_nhwc_bf16(w, "w");
+  const int Nb = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int K = w.size(0), R = w.size(2), S = w.size(3);
+  TORCH_CHECK(w.size(1) == C, "channel mismatch");
+  const int P = (H + 2 * (int)pad - R) / (int)stride + 1;
+  const int Q = (W + 2 * (int)pad - S) / (int)stride + 1;
+  auto out = at::empty({Nb, K, P, Q},
+                       x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  ConvDims d{Nb, P, Q, H, W, C, R, S, (int)stride, (int)pad,
+             Nb * P * Q, K, R * S * C};
+  const dim3 grid((d.N + BN - 1) / BN, (d.M + BM - 1) / BM);
+  const bool fast = (C % BK == 0);
+  auto* kern = fast ? conv_igemm_kernel<0, true> : conv_igemm_kernel<0, false>;
+  hipLaunchKernelGGL(kern, grid, dim3(256), 0, conv_stream(),
+                     bf16_ptr(x), bf16_ptr(w),
+                     reinterpret_cast<__bf16*>(out.data_ptr()), d);
+  return out;
+}
+
+// dx (N,C,H,W) channels_last <- dy (N,K,P,Q) channels_last,
+// wT (C,R,S,K) CONTIGUOUS with taps 180°-rotated (built by the Python side).
+at::Tensor conv_dgrad_igemm(at::Tensor dy, at::Tensor wT, long H, long W,
+                            long stride, long pad) {
+  check_nhwc_bf16(dy, "dy");
+  TORCH_CHECK(wT.is_cuda() && wT.scalar_type() == at::ScalarType::BFloat16 &&
+              wT.is_contiguous(), "wT must be contiguous bf16");
+  const int Nb = dy.size(0), K = dy.size(1), P = dy.size(2), Q = dy.size(3);
+  const int C = wT.size(0), R = wT.size(1), S = wT.size(2);
+  TORCH_CHECK(wT.size(3) == K, "channel mismatch");
+  auto dx = at::empty({Nb, C, H, W},
+                      dy.options().memory_format(at::MemoryFormat::ChannelsLast));
+  ConvDims d{Nb, (int)H, (int)W, P, Q, K, R, S, (int)stride, (int)pad,
+             Nb * (int)H * (int)W, C, R * S * K};
+  const dim3 grid((d.N + BN - 1) / BN, (d.M + BM - 1) / BM);
+  const bool fast = (K % BK == 0);
+  auto* kern = fast ? conv_igemm_kernel<1, true> : conv_igemm_kernel<1, false>;
+  hipLaunchKernelGGL(kern, grid, dim3(256), 0, conv_stream(),
+                     bf16_ptr(dy), bf16_ptr(wT),
+                     reinterpret_cast<__bf16*>(dx.data_ptr()), d);
+  return dx;
+}
+
+// dw fp32 [K][R*S*C] (= channels_last layout of (K,C,R,S)) <- dy, x.
+at::Tensor conv_wgrad_igemm(at::Tensor dy, at::Tensor x, long R, long S,
+                            long stride, long pad) {
+  check_nhwc_bf16(dy, "dy"); check_nhwc_bf16(x, "x");
+  const int Nb = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int K = dy.size(1), P = dy.size(2), Q = dy.size(3);
+  auto dw = at::zeros({K, (long)(R * S * C)}, x.options().dtype(at::kFloat));
+  ConvDims d{Nb, P, Q, H, W, C, (int)R, (int)S, (int)stride, (int)pad,
+             K, (int)(R * S * C), Nb * P * Q};
+  const int tm = (d.M + 63) / 64, tn = (d.N + 63) / 64;
+  const int nchunks = (d.K + BK - 1) / BK;
+  int splits = 768 / (tm * tn);
+  splits = std::max(1, std::min(splits, nchunks));
+  const dim3 grid(tn, tm, splits);
+  const bool fast = (C % BK == 0) && (K % 16 == 0);
+  auto* kern = fast ? conv_wgrad_kernel<true> : conv_wgrad_kernel<false>;
+  hipLaunchKernelGGL(kern, grid, dim3(256), 0, conv_stream(),
+                     bf16_ptr(dy), bf16_ptr(x), dw.data_ptr<float>(), d);
+  return dw;
+}

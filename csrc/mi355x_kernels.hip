@@ -647,7 +647,20 @@ at::Tensor multi_tensor_unscale(std::vector<at::Tensor> grads, double inv_scale)
   return found;
 }
 
+// csrc/conv_igemm.hip — implicit-GEMM MFMA convolution (NHWC bf16)
+at::Tensor conv_fwd_igemm(at::Tensor x, at::Tensor w, long stride, long pad);
+at::Tensor conv_dgrad_igemm(at::Tensor dy, at::Tensor wT, long H, long W,
+                            long stride, long pad);
+at::Tensor conv_wgrad_igemm(at::Tensor dy, at::Tensor x, long R, long S,
+                            long stride, long pad);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("conv_fwd_igemm", &conv_fwd_igemm,
+        "implicit-GEMM conv forward (NHWC bf16, MFMA)");
+  m.def("conv_dgrad_igemm", &conv_dgrad_igemm,
+        "implicit-GEMM conv input-grad (NHWC bf16, MFMA)");
+  m.def("conv_wgrad_igemm", &conv_wgrad_igemm,
+        "implicit-GEMM conv weight-grad (NHWC bf16, MFMA, split-K fp32)");
   m.def("bn_stats", &bn_stats, "per-channel sum/sqsum (NCHW)");
   m.def("bn_fwd", &bn_fwd, "fused BN(+add)(+relu) forward");
   m.def("bn_bwd_reduce", &bn_bwd_reduce, "BN backward reductions w/ relu mask");

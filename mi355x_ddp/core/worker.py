@@ -44,6 +44,9 @@ def build_training(cfg: TrainConfig, device: torch.device, world_size: int,
                    rank: int, distributed: bool = True, wrap: str = "flat"):
     """Model + criterion + optimizer + scheduler + scaler, wrapped for DP mode."""
     model = build_model(cfg.arch, cfg.num_classes).to(device)
+    if cfg.native_ops:
+        from ..ops import MI355Conv2d
+        model = MI355Conv2d.convert(model)
     if cfg.channels_last:
         model = model.to(memory_format=torch.channels_last)
     if distributed and world_size > 1 and cfg.sync_bn:

@@ -131,3 +131,28 @@ def test_scaler_unscale_divides():
     g = torch.full((8,), 4.0)
     scaler.unscale_([g])
     assert torch.allclose(g, torch.ones(8))
+
+
+def test_conv_cpu_fallback_matches_torch():
+    import torch.nn.functional as F
+    from mi355x_ddp.ops import conv2d
+    torch.manual_seed(0)
+    x = torch.randn(2, 8, 16, 16)
+    w = torch.randn(12, 8, 3, 3)
+    got = conv2d(x, w, (1, 1), (1, 1))
+    assert torch.allclose(got, F.conv2d(x, w, None, 1, 1))
+
+
+def test_mi355conv2d_convert_preserves_params():
+    from mi355x_ddp.models import resnet18
+    from mi355x_ddp.ops import MI355Conv2d
+    torch.manual_seed(0)
+    m = resnet18()
+    ref = {k: v.clone() for k, v in m.state_dict().items()}
+    m2 = MI355Conv2d.convert(m)
+    convs = [mod for mod in m2.modules() if isinstance(mod, MI355Conv2d)]
+    assert len(convs) == 20  # 17 convs in blocks+stem, 3 downsample 1x1
+    for k, v in m2.state_dict().items():
+        assert torch.equal(v, ref[k])
+    x = torch.randn(2, 3, 32, 32)
+    assert m2(x).shape == (2, 100)
