@@ -364,6 +364,70 @@ __global__ void multi_tensor_unscale_kernel(MTGradList tl, float inv_scale,
   if (__any(bad) && (threadIdx.x & (WAVE - 1)) == 0) atomicOr(found_inf, 1);
 }
 
+// ---------------------------------------------------------------------------
+// Global average pool (AdaptiveAvgPool2d((1,1)), reference utils/model.py:76).
+// NHWC: one thread per channel, strided over pixels (coalesced across lanes);
+// NCHW: one block per (n,c) row, wave reduction.
+// ---------------------------------------------------------------------------
+template <typename scalar_t>
+__global__ void gap_fwd_nhwc_kernel(const scalar_t* __restrict__ x,
+                                    scalar_t* __restrict__ y,
+                                    int C, int S) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  const int n = blockIdx.y;
+  if (c >= C) return;
+  float s = 0.f;
+  const scalar_t* base = x + (long)n * S * C + c;
+  for (int p = 0; p < S; ++p) s += (float)base[(long)p * C];
+  y[(long)n * C + c] = (scalar_t)(s / S);
+}
+
+template <typename scalar_t>
+__global__ void gap_fwd_nchw_kernel(const scalar_t* __restrict__ x,
+                                    scalar_t* __restrict__ y,
+                                    int C, int S) {
+  __shared__ float lds[32];
+  const long row = blockIdx.x;  // n*C + c
+  float s = 0.f;
+  const scalar_t* base = x + row * S;
+  for (int p = threadIdx.x; p < S; p += blockDim.x) s += (float)base[p];
+  s = block_reduce_sum(s, lds);
+  if (threadIdx.x == 0) y[row] = (scalar_t)(s / S);
+}
+
+template <typename scalar_t, bool CLAST>
+__global__ void gap_bwd_kernel(const scalar_t* __restrict__ dy,
+                               scalar_t* __restrict__ dx,
+                               long total, int C, int S) {
+  const float inv = 1.f / S;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const long n = CLAST ? i / ((long)S * C) : i / ((long)C * S);
+    const int c = CLAST ? (int)(i % C) : (int)((i / S) % C);
+    dx[i] = (scalar_t)((float)dy[n * C + c] * inv);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Per-row rank of the target class: rank = #{j : logit[j] > logit[target]}.
+// acc@k = mean(rank < k) — equivalent to the reference's topk/eq pipeline
+// (utils/util.py:50-64) for distinct logits, one pass, no sort.
+// ---------------------------------------------------------------------------
+template <typename scalar_t>
+__global__ void class_rank_kernel(const scalar_t* __restrict__ logits,
+                                  const long* __restrict__ target,
+                                  int* __restrict__ rank, int N, int C) {
+  __shared__ float lds[32];
+  const int row = blockIdx.x;
+  const scalar_t* xr = logits + (long)row * C;
+  const float tv = (float)xr[target[row]];
+  float cnt = 0.f;
+  for (int j = threadIdx.x; j < C; j += blockDim.x)
+    cnt += ((float)xr[j] > tv) ? 1.f : 0.f;
+  cnt = block_reduce_sum(cnt, lds);
+  if (threadIdx.x == 0) rank[row] = (int)cnt;
+}
+
 int split_for(long per_channel_elems, int nchannels) {
   // enough blocks to fill 256 CUs even for small C
   long want = (2048 + nchannels - 1) / nchannels;
@@ -647,6 +711,67 @@ at::Tensor multi_tensor_unscale(std::vector<at::Tensor> grads, double inv_scale)
   return found;
 }
 
+at::Tensor gap_fwd(at::Tensor x) {
+  CHECK_CUDA(x); check_dense(x, "x");
+  const int N = x.size(0), C = x.size(1);
+  const int S = x.numel() / ((long)N * C);
+  auto y = at::empty({N, C, 1, 1},
+                     is_clast(x)
+                         ? x.options().memory_format(at::MemoryFormat::ChannelsLast)
+                         : x.options());
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::Half, at::ScalarType::BFloat16,
+      x.scalar_type(), "gap_fwd", [&] {
+    if (is_clast(x)) {
+      hipLaunchKernelGGL(gap_fwd_nhwc_kernel<scalar_t>,
+                         dim3((C + 255) / 256, N), dim3(256), 0, cur_stream(),
+                         x.data_ptr<scalar_t>(), y.data_ptr<scalar_t>(), C, S);
+    } else {
+      hipLaunchKernelGGL(gap_fwd_nchw_kernel<scalar_t>, dim3((long)N * C),
+                         dim3(256), 0, cur_stream(),
+                         x.data_ptr<scalar_t>(), y.data_ptr<scalar_t>(), C, S);
+    }
+  });
+  return y;
+}
+
+at::Tensor gap_bwd(at::Tensor dy, at::Tensor x_like) {
+  CHECK_CUDA(dy);
+  const int N = x_like.size(0), C = x_like.size(1);
+  const int S = x_like.numel() / ((long)N * C);
+  const long total = x_like.numel();
+  auto fmt = is_clast(x_like) ? at::MemoryFormat::ChannelsLast
+                              : at::MemoryFormat::Contiguous;
+  auto dx = at::empty_like(x_like, x_like.options().memory_format(fmt));
+  auto dyc = dy.reshape({N, C}).contiguous();
+  const int blocks = (int)std::min((total + 255) / 256, (long)4096);
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::Half, at::ScalarType::BFloat16,
+      dy.scalar_type(), "gap_bwd", [&] {
+    auto launch = [&](auto cl_c) {
+      hipLaunchKernelGGL((gap_bwd_kernel<scalar_t, decltype(cl_c)::value>),
+                         dim3(blocks), dim3(256), 0, cur_stream(),
+                         dyc.data_ptr<scalar_t>(), dx.data_ptr<scalar_t>(),
+                         total, C, S);
+    };
+    is_clast(x_like) ? launch(std::true_type{}) : launch(std::false_type{});
+  });
+  return dx;
+}
+
+at::Tensor class_rank(at::Tensor logits, at::Tensor target) {
+  CHECK_CUDA(logits); CHECK_CONTIG(logits);
+  TORCH_CHECK(target.scalar_type() == at::kLong, "target must be int64");
+  const int N = logits.size(0), C = logits.size(1);
+  auto rank = at::empty({N}, logits.options().dtype(at::kInt));
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::Half, at::ScalarType::BFloat16,
+      logits.scalar_type(), "class_rank", [&] {
+    hipLaunchKernelGGL(class_rank_kernel<scalar_t>, dim3(N),
+                       dim3(C <= 128 ? 64 : 256), 0, cur_stream(),
+                       logits.data_ptr<scalar_t>(), target.data_ptr<long>(),
+                       rank.data_ptr<int>(), N, C);
+  });
+  return rank;
+}
+
 // csrc/conv_igemm.hip — implicit-GEMM MFMA convolution (NHWC bf16)
 at::Tensor conv_fwd_igemm(at::Tensor x, at::Tensor w, long stride, long pad);
 at::Tensor conv_dgrad_igemm(at::Tensor dy, at::Tensor wT, long H, long W,
@@ -670,4 +795,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_tensor_sgd", &multi_tensor_sgd, "fused multi-tensor SGD step");
   m.def("multi_tensor_unscale", &multi_tensor_unscale,
         "multi-tensor grad unscale + inf/nan check");
+  m.def("gap_fwd", &gap_fwd, "global average pool forward (NCHW/NHWC)");
+  m.def("gap_bwd", &gap_bwd, "global average pool backward");
+  m.def("class_rank", &class_rank, "per-row rank of target class (for top-k)");
 }

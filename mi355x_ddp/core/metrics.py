@@ -62,7 +62,16 @@ class ProgressMeter:
 @torch.no_grad()
 def accuracy(output: torch.Tensor, target: torch.Tensor,
              topk: Iterable[int] = (1, 5)) -> list[torch.Tensor]:
-    """Top-k accuracy in percent (reference utils/util.py:50-64)."""
+    """Top-k accuracy in percent (reference utils/util.py:50-64).
+
+    On GPU this runs the native class_rank kernel: one pass computing
+    rank(target) per row — acc@k = mean(rank < k) — instead of ATen's
+    sort-based topk + eq + k reductions.
+    """
+    from ..ops import _backend
+    if _backend.native_enabled(output):
+        rank = _backend.C().class_rank(output.contiguous(), target)
+        return [(rank < k).float().mean().mul(100.0).reshape(1) for k in topk]
     maxk = max(topk)
     batch_size = target.size(0)
     _, pred = output.topk(maxk, 1, True, True)

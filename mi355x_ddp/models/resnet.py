@@ -16,6 +16,7 @@ import torch
 import torch.nn as nn
 
 from ..ops import bn_relu, bn_add_relu
+from ..ops.pool import GlobalAvgPool2d
 
 
 def _conv3x3(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
@@ -95,7 +96,7 @@ class ResNet(nn.Module):
         self.layer2 = self._make_layer(block, 128, num_blocks[1], 2)
         self.layer3 = self._make_layer(block, 256, num_blocks[2], 2)
         self.layer4 = self._make_layer(block, 512, num_blocks[3], 2)
-        self.avgpool = nn.AdaptiveAvgPool2d((1, 1))
+        self.avgpool = GlobalAvgPool2d()  # native HIP global-avg-pool kernel
         self.fc = nn.Linear(512 * block.expansion, num_classes)
 
     def _make_layer(self, block, cout, blocks, stride):
