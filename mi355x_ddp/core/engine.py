@@ -51,6 +51,8 @@ def train_one_epoch(model, loader, criterion, optimizer, epoch: int,
             break
         images = images.to(device, non_blocking=True)
         labels = labels.to(device, non_blocking=True)
+        if cfg.channels_last:
+            images = images.to(memory_format=torch.channels_last)
         _zero_grads(model, optimizer)
 
         if accu == 1:
@@ -122,6 +124,8 @@ def validate(model, loader, criterion, device, cfg: TrainConfig,
     for i, (images, labels) in enumerate(loader):
         images = images.to(device, non_blocking=True)
         labels = labels.to(device, non_blocking=True)
+        if cfg.channels_last:
+            images = images.to(memory_format=torch.channels_last)
         with autocast_ctx(cfg.amp, device.type):
             outputs = model(images)
             loss = criterion(outputs, labels)

@@ -33,6 +33,10 @@ def _native_ok(x: torch.Tensor, weight: torch.Tensor, stride, padding,
         return False
     if stride[0] != stride[1] or padding[0] != padding[1]:
         return False
+    # only take over when the tensors are already NHWC — in NCHW mode the
+    # MIOpen path stays (no per-call layout conversions)
+    if not x.is_contiguous(memory_format=torch.channels_last):
+        return False
     # square filters only (the ResNet zoo is 3x3 / 1x1)
     return weight.shape[2] == weight.shape[3]
 
