@@ -6,7 +6,11 @@ batch size, bf16 channels_last, and prints a table with speedups. Run on an
 MI355X box; output goes to stdout (redirect into gpurun_out/).
 """
 import argparse
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 import torch.nn.functional as F
