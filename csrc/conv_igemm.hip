@@ -332,6 +332,24 @@ void conv_wgrad_kernel(const __bf16* __restrict__ dy,
     }
 }
 
+// wT[c][i][j][k] = w[k][R-1-i][S-1-j][c] — the 180°-rotated transposed filter
+// the dgrad GEMM consumes, built in ONE kernel (the ATen flip+permute+copy
+// chain was 3 launches per conv backward).
+__global__ void build_wT_kernel(const __bf16* __restrict__ w,
+                                __bf16* __restrict__ wT,
+                                int K, int R, int S, int C) {
+  const long total = (long)K * R * S * C;
+  for (long o = (long)blockIdx.x * blockDim.x + threadIdx.x; o < total;
+       o += (long)gridDim.x * blockDim.x) {
+    const int k = (int)(o % K);
+    long t = o / K;
+    const int j = (int)(t % S); t /= S;
+    const int i = (int)(t % R);
+    const int c = (int)(t / R);
+    wT[o] = w[(((long)k * R + (R - 1 - i)) * S + (S - 1 - j)) * C + c];
+  }
+}
+
 hipStream_t conv_stream() { return c10::hip::getCurrentHIPStream().stream(); }
 
 inline void check_nhwc_bf16(const at::Tensor& t, const char* name) {
@@ -348,6 +366,19 @@ inline const __bf16* bf16_ptr(const at::Tensor& t) {
 }
 
 }  // namespace
+
+at::Tensor conv_build_wT(at::Tensor w) {
+  check_nhwc_bf16(w, "w");
+  const int K = w.size(0), C = w.size(1), R = w.size(2), S = w.size(3);
+  auto wT = at::empty({C, R, S, K}, w.options().memory_format(
+                                        at::MemoryFormat::Contiguous));
+  const long total = (long)K * R * S * C;
+  const int blocks = (int)std::min((total + 255) / 256, (long)2048);
+  hipLaunchKernelGGL(build_wT_kernel, dim3(blocks), dim3(256), 0,
+                     conv_stream(), bf16_ptr(w),
+                     reinterpret_cast<__bf16*>(wT.data_ptr()), K, R, S, C);
+  return wT;
+}
 
 // out (N,K,P,Q) channels_last <- x (N,C,H,W) channels_last, w (K,C,R,S)
 // channels_last (memory [K][R][S][C]).

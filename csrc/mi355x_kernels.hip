@@ -205,6 +205,117 @@ __global__ void bn_stats_nhwc_kernel(const scalar_t* __restrict__ x,
   atomicAdd(&sqsum[c], sq);
 }
 
+// v2: full 256-thread utilisation at any C. Threads map to (channel, row):
+// cw = min(C,256) channels x rpb = 256/cw rows per iteration; each block
+// writes its own slice of partial[split][2C] — no atomics, no zero-init.
+// A tiny reduce kernel folds the split axis into the packed [2C] vector the
+// SyncBN all_reduce (and bn_finalize) consume.
+template <typename scalar_t>
+__global__ void bn_stats_nhwc_v2_kernel(const scalar_t* __restrict__ x,
+                                        float* __restrict__ partial,  // [split][2C]
+                                        long P, int C, int cw) {
+  __shared__ float lds[512];
+  const int rpb = blockDim.x / cw;           // rows per iteration
+  const int c_loc = threadIdx.x % cw;
+  const int r = threadIdx.x / cw;
+  const int c = blockIdx.x * cw + c_loc;
+  float s = 0.f, sq = 0.f;
+  if (c < C) {
+    for (long p = (long)blockIdx.y * rpb + r; p < P;
+         p += (long)gridDim.y * rpb) {
+      const float v = (float)x[p * C + c];
+      s += v;
+      sq += v * v;
+    }
+  }
+  // fold the row axis: threads {r=0..rpb-1} with the same c_loc
+  lds[threadIdx.x] = s;
+  lds[256 + threadIdx.x] = sq;
+  __syncthreads();
+  if (r == 0 && c < C) {
+    for (int rr = 1; rr < rpb; ++rr) {
+      s += lds[rr * cw + c_loc];
+      sq += lds[256 + rr * cw + c_loc];
+    }
+    float* row = partial + (long)blockIdx.y * 2 * C;
+    row[c] = s;
+    row[C + c] = sq;
+  }
+}
+
+// partial [split][2C] -> packed [2C]
+__global__ void bn_reduce_partials_kernel(const float* __restrict__ partial,
+                                          float* __restrict__ packed,
+                                          int split, int twoC) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= twoC) return;
+  float s = 0.f;
+  for (int j = 0; j < split; ++j) s += partial[(long)j * twoC + i];
+  packed[i] = s;
+}
+
+// packed {sum, sqsum} + count -> mean, invstd (+ running stats update).
+// Replaces the ~10 ATen launches of eager mean/var/rsqrt/lerp per BN layer.
+__global__ void bn_finalize_kernel(const float* __restrict__ packed,
+                                   float* __restrict__ mean_out,
+                                   float* __restrict__ invstd_out,
+                                   float* __restrict__ running_mean,
+                                   float* __restrict__ running_var,
+                                   float count, float momentum, float eps,
+                                   int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float mean = packed[c] / count;
+  float var = packed[C + c] / count - mean * mean;
+  var = fmaxf(var, 0.f);
+  mean_out[c] = mean;
+  invstd_out[c] = rsqrtf(var + eps);
+  if (running_mean != nullptr) {
+    const float unbiased = var * (count / fmaxf(count - 1.f, 1.f));
+    running_mean[c] += momentum * (mean - running_mean[c]);
+    running_var[c] += momentum * (unbiased - running_var[c]);
+  }
+}
+
+template <typename scalar_t, bool RELU>
+__global__ void bn_bwd_reduce_nhwc_v2_kernel(const scalar_t* __restrict__ dy,
+                                             const scalar_t* __restrict__ x,
+                                             const scalar_t* __restrict__ y,
+                                             const float* __restrict__ mean,
+                                             const float* __restrict__ invstd,
+                                             float* __restrict__ partial,  // [split][2C]
+                                             long P, int C, int cw) {
+  __shared__ float lds[512];
+  const int rpb = blockDim.x / cw;
+  const int c_loc = threadIdx.x % cw;
+  const int r = threadIdx.x / cw;
+  const int c = blockIdx.x * cw + c_loc;
+  float sdy = 0.f, sdyx = 0.f;
+  if (c < C) {
+    const float mu = mean[c], is = invstd[c];
+    for (long p = (long)blockIdx.y * rpb + r; p < P;
+         p += (long)gridDim.y * rpb) {
+      const long idx = p * C + c;
+      float g = (float)dy[idx];
+      if (RELU && (float)y[idx] <= 0.f) g = 0.f;
+      sdy += g;
+      sdyx += g * ((float)x[idx] - mu) * is;
+    }
+  }
+  lds[threadIdx.x] = sdy;
+  lds[256 + threadIdx.x] = sdyx;
+  __syncthreads();
+  if (r == 0 && c < C) {
+    for (int rr = 1; rr < rpb; ++rr) {
+      sdy += lds[rr * cw + c_loc];
+      sdyx += lds[256 + rr * cw + c_loc];
+    }
+    float* row = partial + (long)blockIdx.y * 2 * C;
+    row[c] = sdy;
+    row[C + c] = sdyx;
+  }
+}
+
 template <typename scalar_t, bool RELU>
 __global__ void bn_bwd_reduce_nhwc_kernel(const scalar_t* __restrict__ dy,
                                           const scalar_t* __restrict__ x,
@@ -296,7 +407,7 @@ __global__ void xent_bwd_kernel(const scalar_t* __restrict__ logits,
 // each block linear-scans the per-tensor chunk counts to find its tensor.
 // ---------------------------------------------------------------------------
 constexpr int MT_MAX_TENSORS = 32;
-constexpr int MT_CHUNK = 1 << 16;
+constexpr int MT_CHUNK = 1 << 13;
 
 struct MTTensorList {
   float* p[MT_MAX_TENSORS];
@@ -483,6 +594,132 @@ std::vector<at::Tensor> bn_stats(at::Tensor x) {
     }
   });
   return {sum, sqsum};
+}
+
+namespace {
+// cw = channels handled per block row-group (must divide blockDim 256)
+inline int cw_for(int C) {
+  return (C <= 256 && 256 % C == 0) ? C : 256;
+}
+inline int split_for_nhwc(long P, int cblocks, int rpb) {
+  long want = std::max(1L, 512L / cblocks);
+  long avail = (P + rpb - 1) / rpb;
+  return (int)std::min((long)64, std::min(want, avail));
+}
+}  // namespace
+
+// Per-channel {sum, sqsum} packed into ONE [2C] fp32 tensor (the layout the
+// SyncBN all_reduce and bn_finalize consume).
+at::Tensor bn_stats_packed(at::Tensor x) {
+  CHECK_CUDA(x); check_dense(x, "x");
+  const int N = x.size(0), C = x.size(1);
+  const long S = x.numel() / ((long)N * C);
+  auto opts = x.options().dtype(at::kFloat);
+  at::Tensor result;
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::Half, at::ScalarType::BFloat16,
+      x.scalar_type(), "bn_stats_packed", [&] {
+    if (is_clast(x)) {
+      const long P = (long)N * S;
+      const int cw = cw_for(C), rpb = 256 / cw;
+      const int cblocks = (C + cw - 1) / cw;
+      const int split = split_for_nhwc(P, cblocks, rpb);
+      auto partial = at::empty({split, 2L * C}, opts);
+      hipLaunchKernelGGL(bn_stats_nhwc_v2_kernel<scalar_t>,
+                         dim3(cblocks, split), dim3(256), 0, cur_stream(),
+                         x.data_ptr<scalar_t>(), partial.data_ptr<float>(),
+                         P, C, cw);
+      auto packed = at::empty({2L * C}, opts);
+      hipLaunchKernelGGL(bn_reduce_partials_kernel,
+                         dim3((2 * C + 255) / 256), dim3(256), 0, cur_stream(),
+                         partial.data_ptr<float>(), packed.data_ptr<float>(),
+                         split, 2 * C);
+      // keep `partial` alive until the kernel ran (stream-ordered free is
+      // safe with the caching allocator)
+      (void)partial;
+      // move result out through the captured variable below
+      result = packed;
+    } else {
+      auto packed = at::zeros({2L * C}, opts);
+      const int split = split_for((long)N * S, C);
+      hipLaunchKernelGGL(bn_stats_kernel<scalar_t>, dim3(C, split), dim3(256),
+                         0, cur_stream(),
+                         x.data_ptr<scalar_t>(), packed.data_ptr<float>(),
+                         packed.data_ptr<float>() + C, N, C, (int)S);
+      result = packed;
+    }
+  });
+  return result;
+}
+
+// packed {sum,sqsum} -> (mean, invstd); updates running stats in-place when
+// given. One launch replaces the eager mean/var/clamp/rsqrt/lerp chain.
+std::vector<at::Tensor> bn_finalize(at::Tensor packed, double count,
+                                    double momentum, double eps,
+                                    c10::optional<at::Tensor> running_mean,
+                                    c10::optional<at::Tensor> running_var) {
+  CHECK_CUDA(packed); CHECK_CONTIG(packed);
+  const int C = packed.numel() / 2;
+  auto mean = at::empty({C}, packed.options());
+  auto invstd = at::empty({C}, packed.options());
+  float* rm = running_mean ? running_mean->data_ptr<float>() : nullptr;
+  float* rv = running_var ? running_var->data_ptr<float>() : nullptr;
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0,
+                     cur_stream(), packed.data_ptr<float>(),
+                     mean.data_ptr<float>(), invstd.data_ptr<float>(), rm, rv,
+                     (float)count, (float)momentum, (float)eps, C);
+  return {mean, invstd};
+}
+
+// {sum_dy, sum_dy_xhat} packed [2C], relu mask folded in.
+at::Tensor bn_bwd_reduce_packed(at::Tensor dy, at::Tensor x, at::Tensor mean,
+                                at::Tensor invstd, at::Tensor y, bool relu) {
+  CHECK_CUDA(dy); check_dense(dy, "dy"); check_dense(x, "x");
+  TORCH_CHECK(is_clast(dy) == is_clast(x), "dy/x layout mismatch");
+  const int N = x.size(0), C = x.size(1);
+  const long S = x.numel() / ((long)N * C);
+  auto opts = x.options().dtype(at::kFloat);
+  at::Tensor result;
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::Half, at::ScalarType::BFloat16,
+      x.scalar_type(), "bn_bwd_reduce_packed", [&] {
+    if (is_clast(x)) {
+      const long P = (long)N * S;
+      const int cw = cw_for(C), rpb = 256 / cw;
+      const int cblocks = (C + cw - 1) / cw;
+      const int split = split_for_nhwc(P, cblocks, rpb);
+      auto partial = at::empty({split, 2L * C}, opts);
+      auto launch = [&](auto relu_c) {
+        hipLaunchKernelGGL((bn_bwd_reduce_nhwc_v2_kernel<scalar_t,
+                                                         decltype(relu_c)::value>),
+                           dim3(cblocks, split), dim3(256), 0, cur_stream(),
+                           dy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
+                           y.data_ptr<scalar_t>(), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), partial.data_ptr<float>(),
+                           P, C, cw);
+      };
+      relu ? launch(std::true_type{}) : launch(std::false_type{});
+      auto packed = at::empty({2L * C}, opts);
+      hipLaunchKernelGGL(bn_reduce_partials_kernel,
+                         dim3((2 * C + 255) / 256), dim3(256), 0, cur_stream(),
+                         partial.data_ptr<float>(), packed.data_ptr<float>(),
+                         split, 2 * C);
+      result = packed;
+    } else {
+      auto packed = at::zeros({2L * C}, opts);
+      const int split = split_for((long)N * S, C);
+      auto launch = [&](auto relu_c) {
+        hipLaunchKernelGGL((bn_bwd_reduce_kernel<scalar_t,
+                                                 decltype(relu_c)::value>),
+                           dim3(C, split), dim3(256), 0, cur_stream(),
+                           dy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
+                           y.data_ptr<scalar_t>(), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), packed.data_ptr<float>(),
+                           packed.data_ptr<float>() + C, N, C, (int)S);
+      };
+      relu ? launch(std::true_type{}) : launch(std::false_type{});
+      result = packed;
+    }
+  });
+  return result;
 }
 
 at::Tensor bn_fwd(at::Tensor x, at::Tensor weight, at::Tensor bias,
@@ -773,6 +1010,7 @@ at::Tensor class_rank(at::Tensor logits, at::Tensor target) {
 }
 
 // csrc/conv_igemm.hip — implicit-GEMM MFMA convolution (NHWC bf16)
+at::Tensor conv_build_wT(at::Tensor w);
 at::Tensor conv_fwd_igemm(at::Tensor x, at::Tensor w, long stride, long pad);
 at::Tensor conv_dgrad_igemm(at::Tensor dy, at::Tensor wT, long H, long W,
                             long stride, long pad);
@@ -780,6 +1018,8 @@ at::Tensor conv_wgrad_igemm(at::Tensor dy, at::Tensor x, long R, long S,
                             long stride, long pad);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("conv_build_wT", &conv_build_wT,
+        "build rotated/transposed filter for dgrad (one launch)");
   m.def("conv_fwd_igemm", &conv_fwd_igemm,
         "implicit-GEMM conv forward (NHWC bf16, MFMA)");
   m.def("conv_dgrad_igemm", &conv_dgrad_igemm,
@@ -787,6 +1027,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_wgrad_igemm", &conv_wgrad_igemm,
         "implicit-GEMM conv weight-grad (NHWC bf16, MFMA, split-K fp32)");
   m.def("bn_stats", &bn_stats, "per-channel sum/sqsum (NCHW)");
+  m.def("bn_stats_packed", &bn_stats_packed,
+        "per-channel {sum,sqsum} packed [2C], no-atomic NHWC v2");
+  m.def("bn_finalize", &bn_finalize,
+        "packed stats + count -> mean/invstd (+running update), one launch");
+  m.def("bn_bwd_reduce_packed", &bn_bwd_reduce_packed,
+        "BN backward reductions packed [2C] w/ relu mask");
   m.def("bn_fwd", &bn_fwd, "fused BN(+add)(+relu) forward");
   m.def("bn_bwd_reduce", &bn_bwd_reduce, "BN backward reductions w/ relu mask");
   m.def("bn_bwd", &bn_bwd, "BN backward apply");

@@ -94,25 +94,39 @@ class _FusedBNFunction(torch.autograd.Function):
         ctx.fmt = fmt
         N, C, H, W = x.shape
         if training:
-            s, sq = _stats(x)
             # count is computed arithmetically (equal per-rank batches are
             # guaranteed by drop_last sharding) — no .item() host sync per BN
             # layer, which also keeps the op hipGraph-capturable.
             cnt = float(N * H * W)
-            if process_group is not None and dist.is_initialized() \
-                    and dist.get_world_size(process_group) > 1:
-                packed = torch.cat([s, sq])
-                dist.all_reduce(packed, op=dist.ReduceOp.SUM, group=process_group)
-                s, sq = packed[:C], packed[C:2 * C]
+            synced = process_group is not None and dist.is_initialized() \
+                and dist.get_world_size(process_group) > 1
+            if synced:
                 cnt *= dist.get_world_size(process_group)
-            mean = s / cnt
-            var = sq / cnt - mean * mean
-            var = var.clamp_min_(0.0)
-            invstd = torch.rsqrt(var + eps)
-            if running_mean is not None:
-                unbiased = var * (cnt / max(cnt - 1.0, 1.0))
-                running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
-                running_var.mul_(1 - momentum).add_(unbiased, alpha=momentum)
+            if _backend.native_enabled(x):
+                # packed {sum,sqsum} [2C]: one stats kernel, one optional RCCL
+                # all_reduce, one finalize launch (mean/invstd + running-stats
+                # update) — replaces the eager mean/var/rsqrt/lerp chain.
+                packed = _backend.C().bn_stats_packed(x)
+                if synced:
+                    dist.all_reduce(packed, op=dist.ReduceOp.SUM,
+                                    group=process_group)
+                mean, invstd = _backend.C().bn_finalize(
+                    packed, cnt, momentum, eps, running_mean, running_var)
+            else:
+                s, sq = _stats(x)
+                if synced:
+                    packed = torch.cat([s, sq])
+                    dist.all_reduce(packed, op=dist.ReduceOp.SUM,
+                                    group=process_group)
+                    s, sq = packed[:C], packed[C:2 * C]
+                mean = s / cnt
+                var = sq / cnt - mean * mean
+                var = var.clamp_min_(0.0)
+                invstd = torch.rsqrt(var + eps)
+                if running_mean is not None:
+                    unbiased = var * (cnt / max(cnt - 1.0, 1.0))
+                    running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
+                    running_var.mul_(1 - momentum).add_(unbiased, alpha=momentum)
         else:
             mean = running_mean.float()
             invstd = torch.rsqrt(running_var.float() + eps)
@@ -130,19 +144,27 @@ class _FusedBNFunction(torch.autograd.Function):
     def backward(ctx, dy):
         x, weight, mean, invstd, y = ctx.saved_tensors
         dy = dy.contiguous(memory_format=ctx.fmt)
-        sum_dy, sum_dy_xhat = _bwd_reduce(dy, x, mean, invstd, y, ctx.relu)
         count = ctx.count
+        C = x.shape[1]
+        synced = ctx.training and ctx.process_group is not None \
+            and dist.is_initialized() \
+            and dist.get_world_size(ctx.process_group) > 1
+        if _backend.native_enabled(x):
+            packed = _backend.C().bn_bwd_reduce_packed(dy, x, mean, invstd, y,
+                                                       ctx.relu)
+        else:
+            s_dy, s_dyx = _bwd_reduce(dy, x, mean, invstd, y, ctx.relu)
+            packed = torch.cat([s_dy, s_dyx])
         # dgamma/dbeta are the LOCAL sums — the DP gradient all-reduce averages
         # them like every other parameter grad. The cross-rank-summed versions
         # are only for dx (whose formula needs the GLOBAL batch statistics).
-        dgamma = sum_dy_xhat.to(weight.dtype)
-        dbeta = sum_dy.to(weight.dtype)
-        if ctx.training and ctx.process_group is not None and dist.is_initialized() \
-                and dist.get_world_size(ctx.process_group) > 1:
-            C = sum_dy.numel()
-            packed = torch.cat([sum_dy, sum_dy_xhat])
+        # copy=True: the all_reduce below mutates `packed` in place and these
+        # must keep the LOCAL values
+        dgamma = packed[C:].to(weight.dtype, copy=True)
+        dbeta = packed[:C].to(weight.dtype, copy=True)
+        if synced:
             dist.all_reduce(packed, op=dist.ReduceOp.SUM, group=ctx.process_group)
-            sum_dy, sum_dy_xhat = packed[:C], packed[C:]
+        sum_dy, sum_dy_xhat = packed[:C], packed[C:]
         dx, dresidual = _bwd_apply(dy, x, weight, mean, invstd, sum_dy,
                                    sum_dy_xhat, count, y, ctx.relu,
                                    ctx.training, ctx.has_residual)

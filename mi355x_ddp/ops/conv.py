@@ -57,7 +57,7 @@ class _ConvIGEMM(torch.autograd.Function):
         dx = dw = None
         if ctx.needs_input_grad[0]:
             # 180°-rotated, (C,R,S,K)-transposed filter for the dgrad GEMM
-            wT = weight.flip(2, 3).permute(1, 2, 3, 0).contiguous()
+            wT = _backend.C().conv_build_wT(weight)
             dx = _backend.C().conv_dgrad_igemm(dy, wT, x.shape[2], x.shape[3],
                                                ctx.stride, ctx.padding)
         if ctx.needs_input_grad[1]:

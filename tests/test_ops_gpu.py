@@ -276,3 +276,90 @@ def test_fused_sgd_optimizer_gpu_matches_torch():
             o.step()
     for p1, p2 in zip(m1.parameters(), m2.parameters()):
         assert torch.allclose(p1, p2, atol=1e-5)
+
+
+@pytest.mark.parametrize("shape", SHAPES)
+@pytest.mark.parametrize("clast", [False, True])
+def test_bn_stats_packed_and_finalize(shape, clast):
+    from mi355x_ddp.ops import _backend
+    x = torch.randn(*shape, device=DEV, dtype=torch.bfloat16)
+    if clast:
+        x = x.to(memory_format=torch.channels_last)
+    N, C, H, W = shape
+    packed = _backend.C().bn_stats_packed(x)
+    xf = x.float()
+    ref_s = xf.sum(dim=(0, 2, 3))
+    ref_sq = (xf * xf).sum(dim=(0, 2, 3))
+    assert torch.allclose(packed[:C], ref_s, atol=2.0, rtol=1e-2)
+    assert torch.allclose(packed[C:], ref_sq, atol=2.0, rtol=1e-2)
+
+    cnt = float(N * H * W)
+    rm = torch.zeros(C, device=DEV)
+    rv = torch.ones(C, device=DEV)
+    mean, invstd = _backend.C().bn_finalize(packed, cnt, 0.1, 1e-5, rm, rv)
+    ref_mean = ref_s / cnt
+    ref_var = (ref_sq / cnt - ref_mean * ref_mean).clamp_min(0)
+    assert torch.allclose(mean, ref_mean, atol=1e-2, rtol=1e-2)
+    assert torch.allclose(invstd, torch.rsqrt(ref_var + 1e-5), atol=1e-2, rtol=1e-2)
+    unb = ref_var * cnt / (cnt - 1)
+    assert torch.allclose(rm, 0.1 * ref_mean, atol=1e-3, rtol=1e-2)
+    assert torch.allclose(rv, 0.9 + 0.1 * unb, atol=1e-2, rtol=1e-2)
+
+
+@pytest.mark.parametrize("clast", [False, True])
+def test_bn_bwd_reduce_packed(clast):
+    from mi355x_ddp.ops import _backend
+    N, C, H, W = 8, 128, 16, 16
+    fmt = torch.channels_last if clast else torch.contiguous_format
+    x = torch.randn(N, C, H, W, device=DEV, dtype=torch.bfloat16).contiguous(memory_format=fmt)
+    dy = torch.randn_like(x).contiguous(memory_format=fmt)
+    y = torch.randn_like(x).contiguous(memory_format=fmt)
+    mean = x.float().mean(dim=(0, 2, 3)).contiguous()
+    invstd = torch.rand(C, device=DEV) + 0.5
+    packed = _backend.C().bn_bwd_reduce_packed(dy, x, mean, invstd, y, True)
+    g = dy.float() * (y.float() > 0)
+    xhat = (x.float() - mean.view(1, -1, 1, 1)) * invstd.view(1, -1, 1, 1)
+    assert torch.allclose(packed[:C], g.sum(dim=(0, 2, 3)), atol=1.0, rtol=2e-2)
+    assert torch.allclose(packed[C:], (g * xhat).sum(dim=(0, 2, 3)), atol=1.0, rtol=2e-2)
+
+
+@pytest.mark.parametrize("clast", [False, True])
+def test_global_avgpool(clast):
+    from mi355x_ddp.ops.pool import GlobalAvgPool2d
+    x = torch.randn(8, 512, 4, 4, device=DEV, dtype=torch.bfloat16)
+    if clast:
+        x = x.to(memory_format=torch.channels_last)
+    x.requires_grad_(True)
+    y = GlobalAvgPool2d()(x)
+    assert y.shape == (8, 512, 1, 1)
+    ref = x.float().mean(dim=(2, 3), keepdim=True)
+    assert torch.allclose(y.float(), ref, atol=1e-2, rtol=1e-2)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    ref_dx = (dy.float() / 16).expand(8, 512, 4, 4)
+    assert torch.allclose(x.grad.float(), ref_dx, atol=1e-2, rtol=1e-2)
+
+
+def test_class_rank_accuracy():
+    from mi355x_ddp.core.metrics import accuracy
+    torch.manual_seed(3)
+    logits = torch.randn(64, 100, device=DEV)
+    target = torch.randint(0, 100, (64,), device=DEV)
+    got1, got5 = accuracy(logits, target, (1, 5))
+    # reference: topk pipeline on CPU
+    maxk = 5
+    _, pred = logits.topk(maxk, 1, True, True)
+    correct = pred.t().eq(target.view(1, -1))
+    ref1 = correct[:1].float().sum() * 100.0 / 64
+    ref5 = correct[:5].reshape(-1).float().sum() * 100.0 / 64
+    assert torch.allclose(got1.cpu().reshape(()), ref1.cpu(), atol=1e-4)
+    assert torch.allclose(got5.cpu().reshape(()), ref5.cpu(), atol=1e-4)
+
+
+def test_conv_build_wT():
+    from mi355x_ddp.ops import _backend
+    w = torch.randn(64, 128, 3, 3, device=DEV, dtype=torch.bfloat16) \
+        .to(memory_format=torch.channels_last)
+    got = _backend.C().conv_build_wT(w)
+    ref = w.flip(2, 3).permute(1, 2, 3, 0).contiguous()
+    assert torch.equal(got, ref)
