@@ -110,6 +110,37 @@ __global__ void bn_fwd_kernel(const scalar_t* __restrict__ x,
   }
 }
 
+// Vectorized variant: 8 elements (16 B) per thread. Correct when the
+// channel run stays within one c-block: CLAST needs C %% 8 == 0, NCHW needs
+// S %% 8 == 0 (true for every ResNet CIFAR stage).
+template <typename scalar_t, bool RELU, bool HAS_RES, bool CLAST>
+__global__ void bn_fwd_vec_kernel(const scalar_t* __restrict__ x,
+                                  const scalar_t* __restrict__ res,
+                                  scalar_t* __restrict__ y,
+                                  const float* __restrict__ weight,
+                                  const float* __restrict__ bias,
+                                  const float* __restrict__ mean,
+                                  const float* __restrict__ invstd,
+                                  long nvec, int C, int S) {
+  for (long v = (long)blockIdx.x * blockDim.x + threadIdx.x; v < nvec;
+       v += (long)gridDim.x * blockDim.x) {
+    const long i = v * 8;
+    scalar_t x8[8], r8[8], y8[8];
+    *(float4*)x8 = *(const float4*)(x + i);
+    if (HAS_RES) *(float4*)r8 = *(const float4*)(res + i);
+    const int cbase = CLAST ? (int)(i % C) : (int)((i / S) % C);
+    #pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const int c = CLAST ? cbase + e : cbase;
+      float o = ((float)x8[e] - mean[c]) * invstd[c] * weight[c] + bias[c];
+      if (HAS_RES) o += (float)r8[e];
+      if (RELU) o = fmaxf(o, 0.f);
+      y8[e] = (scalar_t)o;
+    }
+    *(float4*)(y + i) = *(float4*)y8;
+  }
+}
+
 // ---------------------------------------------------------------------------
 // BN backward reductions: per-channel sum_dy and sum_dy_xhat with the ReLU
 // mask (y > 0) folded in — the fused-ReLU backward never materialises a mask.
@@ -181,6 +212,47 @@ __global__ void bn_bwd_kernel(const scalar_t* __restrict__ dy,
 }
 
 
+template <typename scalar_t, bool RELU, bool TRAIN, bool NEED_DRES, bool CLAST>
+__global__ void bn_bwd_vec_kernel(const scalar_t* __restrict__ dy,
+                                  const scalar_t* __restrict__ x,
+                                  const scalar_t* __restrict__ y,
+                                  scalar_t* __restrict__ dx,
+                                  scalar_t* __restrict__ dres,
+                                  const float* __restrict__ weight,
+                                  const float* __restrict__ mean,
+                                  const float* __restrict__ invstd,
+                                  const float* __restrict__ sum_dy,
+                                  const float* __restrict__ sum_dy_xhat,
+                                  float inv_count, long nvec, int C, int S) {
+  for (long v = (long)blockIdx.x * blockDim.x + threadIdx.x; v < nvec;
+       v += (long)gridDim.x * blockDim.x) {
+    const long i = v * 8;
+    scalar_t dy8[8], x8[8], y8[8], o8[8], dr8[8];
+    *(float4*)dy8 = *(const float4*)(dy + i);
+    if (TRAIN) *(float4*)x8 = *(const float4*)(x + i);
+    if (RELU) *(float4*)y8 = *(const float4*)(y + i);
+    const int cbase = CLAST ? (int)(i % C) : (int)((i / S) % C);
+    #pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const int c = CLAST ? cbase + e : cbase;
+      float g = (float)dy8[e];
+      if (RELU && (float)y8[e] <= 0.f) g = 0.f;
+      if (NEED_DRES) dr8[e] = (scalar_t)g;
+      const float w_is = weight[c] * invstd[c];
+      float o;
+      if (TRAIN) {
+        const float xhat = ((float)x8[e] - mean[c]) * invstd[c];
+        o = w_is * (g - sum_dy[c] * inv_count - xhat * sum_dy_xhat[c] * inv_count);
+      } else {
+        o = w_is * g;
+      }
+      o8[e] = (scalar_t)o;
+    }
+    *(float4*)(dx + i) = *(float4*)o8;
+    if (NEED_DRES) *(float4*)(dres + i) = *(float4*)dr8;
+  }
+}
+
 // ---------------------------------------------------------------------------
 // NHWC (channels_last) variants — the MI355X-preferred layout: per-pixel
 // channels are contiguous, so stats/reductions coalesce across lanes on the
@@ -205,53 +277,80 @@ __global__ void bn_stats_nhwc_kernel(const scalar_t* __restrict__ x,
   atomicAdd(&sqsum[c], sq);
 }
 
-// v2: full 256-thread utilisation at any C. Threads map to (channel, row):
-// cw = min(C,256) channels x rpb = 256/cw rows per iteration; each block
-// writes its own slice of partial[split][2C] — no atomics, no zero-init.
-// A tiny reduce kernel folds the split axis into the packed [2C] vector the
-// SyncBN all_reduce (and bn_finalize) consume.
+// v2: NHWC stats with full thread utilisation AND 16-byte loads. Each thread
+// owns VEC=8 consecutive channels (one dwordx4 per pixel-row visited), tpr =
+// C/8 threads span the channels, rpb = 256/tpr rows advance per iteration.
+// The row axis folds with an LDS tree; each block writes its slice of
+// partial[split][2C] — no atomics, no zero-init. Requires C % 8 == 0 (every
+// ResNet BN); other C take the v1 atomic kernel.
+typedef __attribute__((ext_vector_type(8))) __bf16 bnbf16x8;
+
 template <typename scalar_t>
 __global__ void bn_stats_nhwc_v2_kernel(const scalar_t* __restrict__ x,
                                         float* __restrict__ partial,  // [split][2C]
-                                        long P, int C, int cw) {
-  __shared__ float lds[512];
-  const int rpb = blockDim.x / cw;           // rows per iteration
-  const int c_loc = threadIdx.x % cw;
-  const int r = threadIdx.x / cw;
-  const int c = blockIdx.x * cw + c_loc;
-  float s = 0.f, sq = 0.f;
-  if (c < C) {
-    for (long p = (long)blockIdx.y * rpb + r; p < P;
-         p += (long)gridDim.y * rpb) {
-      const float v = (float)x[p * C + c];
-      s += v;
-      sq += v * v;
+                                        long P, int C, int tpr) {
+  __shared__ float lds[2 * 256 * 8];
+  const int rpb = blockDim.x / tpr;
+  const int c_idx = threadIdx.x % tpr;
+  const int r = threadIdx.x / tpr;
+  const int c0 = c_idx * 8;
+  float s[8] = {}, sq[8] = {};
+  for (long p = (long)blockIdx.y * rpb + r; p < P;
+       p += (long)gridDim.y * rpb) {
+    scalar_t v8[8];
+    if constexpr (sizeof(scalar_t) == 2) {
+      *(float4*)v8 = *(const float4*)(x + p * C + c0);
+    } else {
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) v8[e] = x[p * C + c0 + e];
+    }
+    #pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const float v = (float)v8[e];
+      s[e] += v;
+      sq[e] += v * v;
     }
   }
-  // fold the row axis: threads {r=0..rpb-1} with the same c_loc
-  lds[threadIdx.x] = s;
-  lds[256 + threadIdx.x] = sq;
-  __syncthreads();
-  if (r == 0 && c < C) {
-    for (int rr = 1; rr < rpb; ++rr) {
-      s += lds[rr * cw + c_loc];
-      sq += lds[256 + rr * cw + c_loc];
+  #pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    lds[threadIdx.x * 8 + e] = s[e];
+    lds[2048 + threadIdx.x * 8 + e] = sq[e];
+  }
+  // fold rows: tree over the r axis (threads r, r+stride share c_idx)
+  for (int stride = rpb >> 1; stride > 0; stride >>= 1) {
+    __syncthreads();
+    if (r < stride) {
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        lds[threadIdx.x * 8 + e] += lds[(threadIdx.x + stride * tpr) * 8 + e];
+        lds[2048 + threadIdx.x * 8 + e] +=
+            lds[2048 + (threadIdx.x + stride * tpr) * 8 + e];
+      }
     }
+  }
+  __syncthreads();
+  if (r == 0) {
     float* row = partial + (long)blockIdx.y * 2 * C;
-    row[c] = s;
-    row[C + c] = sq;
+    #pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      row[c0 + e] = lds[threadIdx.x * 8 + e];
+      row[C + c0 + e] = lds[2048 + threadIdx.x * 8 + e];
+    }
   }
 }
 
-// partial [split][2C] -> packed [2C]
+// partial [split][2C] -> packed [2C]: one block per output element, the
+// split axis reduced by 256 threads (split <= 512 -> <=2 loads per thread).
 __global__ void bn_reduce_partials_kernel(const float* __restrict__ partial,
                                           float* __restrict__ packed,
                                           int split, int twoC) {
-  const int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= twoC) return;
+  __shared__ float lds[32];
+  const int i = blockIdx.x;
   float s = 0.f;
-  for (int j = 0; j < split; ++j) s += partial[(long)j * twoC + i];
-  packed[i] = s;
+  for (int j = threadIdx.x; j < split; j += blockDim.x)
+    s += partial[(long)j * twoC + i];
+  s = block_reduce_sum(s, lds);
+  if (threadIdx.x == 0) packed[i] = s;
 }
 
 // packed {sum, sqsum} + count -> mean, invstd (+ running stats update).
@@ -284,35 +383,67 @@ __global__ void bn_bwd_reduce_nhwc_v2_kernel(const scalar_t* __restrict__ dy,
                                              const float* __restrict__ mean,
                                              const float* __restrict__ invstd,
                                              float* __restrict__ partial,  // [split][2C]
-                                             long P, int C, int cw) {
-  __shared__ float lds[512];
-  const int rpb = blockDim.x / cw;
-  const int c_loc = threadIdx.x % cw;
-  const int r = threadIdx.x / cw;
-  const int c = blockIdx.x * cw + c_loc;
-  float sdy = 0.f, sdyx = 0.f;
-  if (c < C) {
-    const float mu = mean[c], is = invstd[c];
-    for (long p = (long)blockIdx.y * rpb + r; p < P;
-         p += (long)gridDim.y * rpb) {
-      const long idx = p * C + c;
-      float g = (float)dy[idx];
-      if (RELU && (float)y[idx] <= 0.f) g = 0.f;
-      sdy += g;
-      sdyx += g * ((float)x[idx] - mu) * is;
+                                             long P, int C, int tpr) {
+  __shared__ float lds[2 * 256 * 8];
+  const int rpb = blockDim.x / tpr;
+  const int c_idx = threadIdx.x % tpr;
+  const int r = threadIdx.x / tpr;
+  const int c0 = c_idx * 8;
+  float mu[8], is[8];
+  #pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    mu[e] = mean[c0 + e];
+    is[e] = invstd[c0 + e];
+  }
+  float sdy[8] = {}, sdyx[8] = {};
+  for (long p = (long)blockIdx.y * rpb + r; p < P;
+       p += (long)gridDim.y * rpb) {
+    const long base = p * C + c0;
+    scalar_t dy8[8], x8[8], y8[8];
+    if constexpr (sizeof(scalar_t) == 2) {
+      *(float4*)dy8 = *(const float4*)(dy + base);
+      *(float4*)x8 = *(const float4*)(x + base);
+      if (RELU) *(float4*)y8 = *(const float4*)(y + base);
+    } else {
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        dy8[e] = dy[base + e];
+        x8[e] = x[base + e];
+        if (RELU) y8[e] = y[base + e];
+      }
+    }
+    #pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float g = (float)dy8[e];
+      if (RELU && (float)y8[e] <= 0.f) g = 0.f;
+      sdy[e] += g;
+      sdyx[e] += g * ((float)x8[e] - mu[e]) * is[e];
     }
   }
-  lds[threadIdx.x] = sdy;
-  lds[256 + threadIdx.x] = sdyx;
-  __syncthreads();
-  if (r == 0 && c < C) {
-    for (int rr = 1; rr < rpb; ++rr) {
-      sdy += lds[rr * cw + c_loc];
-      sdyx += lds[256 + rr * cw + c_loc];
+  #pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    lds[threadIdx.x * 8 + e] = sdy[e];
+    lds[2048 + threadIdx.x * 8 + e] = sdyx[e];
+  }
+  for (int stride = rpb >> 1; stride > 0; stride >>= 1) {
+    __syncthreads();
+    if (r < stride) {
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        lds[threadIdx.x * 8 + e] += lds[(threadIdx.x + stride * tpr) * 8 + e];
+        lds[2048 + threadIdx.x * 8 + e] +=
+            lds[2048 + (threadIdx.x + stride * tpr) * 8 + e];
+      }
     }
+  }
+  __syncthreads();
+  if (r == 0) {
     float* row = partial + (long)blockIdx.y * 2 * C;
-    row[c] = sdy;
-    row[C + c] = sdyx;
+    #pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      row[c0 + e] = lds[threadIdx.x * 8 + e];
+      row[C + c0 + e] = lds[2048 + threadIdx.x * 8 + e];
+    }
   }
 }
 
@@ -597,14 +728,18 @@ std::vector<at::Tensor> bn_stats(at::Tensor x) {
 }
 
 namespace {
-// cw = channels handled per block row-group (must divide blockDim 256)
-inline int cw_for(int C) {
-  return (C <= 256 && 256 % C == 0) ? C : 256;
+// v2 NHWC kernels: each thread owns 8 consecutive channels, tpr = C/8
+// threads per row, rpb = 256/tpr rows per iteration. Usable when C/8 is a
+// power-of-two divisor of 256 (every ResNet18/50 BN width).
+inline bool v2_ok(int C, int elem_size) {
+  if (elem_size != 2 || C % 8 != 0) return false;
+  const int tpr = C / 8;
+  return tpr <= 256 && 256 % tpr == 0 && (tpr & (tpr - 1)) == 0;
 }
-inline int split_for_nhwc(long P, int cblocks, int rpb) {
-  long want = std::max(1L, 512L / cblocks);
+inline int split_for_nhwc(long P, int rpb) {
+  long want = 512;
   long avail = (P + rpb - 1) / rpb;
-  return (int)std::min((long)64, std::min(want, avail));
+  return (int)std::min((long)512, std::min(want, avail));
 }
 }  // namespace
 
@@ -618,25 +753,30 @@ at::Tensor bn_stats_packed(at::Tensor x) {
   at::Tensor result;
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::Half, at::ScalarType::BFloat16,
       x.scalar_type(), "bn_stats_packed", [&] {
-    if (is_clast(x)) {
+    if (is_clast(x) && v2_ok(C, (int)sizeof(scalar_t))) {
       const long P = (long)N * S;
-      const int cw = cw_for(C), rpb = 256 / cw;
-      const int cblocks = (C + cw - 1) / cw;
-      const int split = split_for_nhwc(P, cblocks, rpb);
+      const int tpr = C / 8, rpb = 256 / tpr;
+      const int split = split_for_nhwc(P, rpb);
       auto partial = at::empty({split, 2L * C}, opts);
       hipLaunchKernelGGL(bn_stats_nhwc_v2_kernel<scalar_t>,
-                         dim3(cblocks, split), dim3(256), 0, cur_stream(),
+                         dim3(1, split), dim3(256), 0, cur_stream(),
                          x.data_ptr<scalar_t>(), partial.data_ptr<float>(),
-                         P, C, cw);
+                         P, C, tpr);
       auto packed = at::empty({2L * C}, opts);
       hipLaunchKernelGGL(bn_reduce_partials_kernel,
-                         dim3((2 * C + 255) / 256), dim3(256), 0, cur_stream(),
+                         dim3(2 * C), dim3(256), 0, cur_stream(),
                          partial.data_ptr<float>(), packed.data_ptr<float>(),
                          split, 2 * C);
-      // keep `partial` alive until the kernel ran (stream-ordered free is
-      // safe with the caching allocator)
-      (void)partial;
-      // move result out through the captured variable below
+      result = packed;
+    } else if (is_clast(x)) {
+      const long P = (long)N * S;
+      auto packed = at::zeros({2L * C}, opts);
+      const int cblocks = (C + 255) / 256;
+      const int split = (int)std::min(P, (long)std::max(1, 768 / cblocks));
+      hipLaunchKernelGGL(bn_stats_nhwc_kernel<scalar_t>, dim3(cblocks, split),
+                         dim3(256), 0, cur_stream(),
+                         x.data_ptr<scalar_t>(), packed.data_ptr<float>(),
+                         packed.data_ptr<float>() + C, P, C);
       result = packed;
     } else {
       auto packed = at::zeros({2L * C}, opts);
@@ -681,27 +821,42 @@ at::Tensor bn_bwd_reduce_packed(at::Tensor dy, at::Tensor x, at::Tensor mean,
   at::Tensor result;
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::Half, at::ScalarType::BFloat16,
       x.scalar_type(), "bn_bwd_reduce_packed", [&] {
-    if (is_clast(x)) {
+    if (is_clast(x) && v2_ok(C, (int)sizeof(scalar_t))) {
       const long P = (long)N * S;
-      const int cw = cw_for(C), rpb = 256 / cw;
-      const int cblocks = (C + cw - 1) / cw;
-      const int split = split_for_nhwc(P, cblocks, rpb);
+      const int tpr = C / 8, rpb = 256 / tpr;
+      const int split = split_for_nhwc(P, rpb);
       auto partial = at::empty({split, 2L * C}, opts);
       auto launch = [&](auto relu_c) {
         hipLaunchKernelGGL((bn_bwd_reduce_nhwc_v2_kernel<scalar_t,
                                                          decltype(relu_c)::value>),
-                           dim3(cblocks, split), dim3(256), 0, cur_stream(),
+                           dim3(1, split), dim3(256), 0, cur_stream(),
                            dy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
                            y.data_ptr<scalar_t>(), mean.data_ptr<float>(),
                            invstd.data_ptr<float>(), partial.data_ptr<float>(),
-                           P, C, cw);
+                           P, C, tpr);
       };
       relu ? launch(std::true_type{}) : launch(std::false_type{});
       auto packed = at::empty({2L * C}, opts);
       hipLaunchKernelGGL(bn_reduce_partials_kernel,
-                         dim3((2 * C + 255) / 256), dim3(256), 0, cur_stream(),
+                         dim3(2 * C), dim3(256), 0, cur_stream(),
                          partial.data_ptr<float>(), packed.data_ptr<float>(),
                          split, 2 * C);
+      result = packed;
+    } else if (is_clast(x)) {
+      const long P = (long)N * S;
+      auto packed = at::zeros({2L * C}, opts);
+      const int cblocks = (C + 255) / 256;
+      const int split = (int)std::min(P, (long)std::max(1, 768 / cblocks));
+      auto launch = [&](auto relu_c) {
+        hipLaunchKernelGGL((bn_bwd_reduce_nhwc_kernel<scalar_t,
+                                                      decltype(relu_c)::value>),
+                           dim3(cblocks, split), dim3(256), 0, cur_stream(),
+                           dy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
+                           y.data_ptr<scalar_t>(), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), packed.data_ptr<float>(),
+                           packed.data_ptr<float>() + C, P, C);
+      };
+      relu ? launch(std::true_type{}) : launch(std::false_type{});
       result = packed;
     } else {
       auto packed = at::zeros({2L * C}, opts);
@@ -738,9 +893,25 @@ at::Tensor bn_fwd(at::Tensor x, at::Tensor weight, at::Tensor bias,
   auto wf = weight.to(at::kFloat).contiguous();
   auto bf = bias.to(at::kFloat).contiguous();
   const int blocks = (int)std::min((total + 1023) / 1024, (long)4096);
+  const bool vec_ok = (total % 8 == 0) &&
+      (clast ? (C % 8 == 0) : (S % 8 == 0));
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::Half, at::ScalarType::BFloat16,
       x.scalar_type(), "bn_fwd", [&] {
     auto launch = [&](auto relu_c, auto res_c, auto cl_c) {
+      if (vec_ok && sizeof(scalar_t) == 2) {
+        const long nvec = total / 8;
+        const int vblocks = (int)std::min((nvec + 255) / 256, (long)2048);
+        hipLaunchKernelGGL((bn_fwd_vec_kernel<scalar_t, decltype(relu_c)::value,
+                                              decltype(res_c)::value,
+                                              decltype(cl_c)::value>),
+                           dim3(vblocks), dim3(256), 0, cur_stream(),
+                           x.data_ptr<scalar_t>(),
+                           has_res ? residual.data_ptr<scalar_t>() : nullptr,
+                           y.data_ptr<scalar_t>(), wf.data_ptr<float>(),
+                           bf.data_ptr<float>(), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), nvec, C, S);
+        return;
+      }
       hipLaunchKernelGGL((bn_fwd_kernel<scalar_t, decltype(relu_c)::value,
                                         decltype(res_c)::value,
                                         decltype(cl_c)::value>),
@@ -822,7 +993,26 @@ std::vector<at::Tensor> bn_bwd(at::Tensor dy, at::Tensor x, at::Tensor weight,
   const int blocks = (int)std::min((total + 1023) / 1024, (long)4096);
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::Half, at::ScalarType::BFloat16,
       x.scalar_type(), "bn_bwd", [&] {
+    const bool vec_ok = (total % 8 == 0) &&
+        (clast ? (C % 8 == 0) : (S % 8 == 0));
     auto launch = [&](auto relu_c, auto train_c, auto dres_c, auto cl_c) {
+      if (vec_ok && sizeof(scalar_t) == 2) {
+        const long nvec = total / 8;
+        const int vblocks = (int)std::min((nvec + 255) / 256, (long)2048);
+        hipLaunchKernelGGL((bn_bwd_vec_kernel<scalar_t, decltype(relu_c)::value,
+                                              decltype(train_c)::value,
+                                              decltype(dres_c)::value,
+                                              decltype(cl_c)::value>),
+                           dim3(vblocks), dim3(256), 0, cur_stream(),
+                           dy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
+                           y.data_ptr<scalar_t>(), dx.data_ptr<scalar_t>(),
+                           need_dres ? dres.data_ptr<scalar_t>() : nullptr,
+                           wf.data_ptr<float>(), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), sum_dy.data_ptr<float>(),
+                           sum_dy_xhat.data_ptr<float>(),
+                           (float)(1.0 / count), nvec, C, S);
+        return;
+      }
       hipLaunchKernelGGL((bn_bwd_kernel<scalar_t, decltype(relu_c)::value,
                                         decltype(train_c)::value,
                                         decltype(dres_c)::value,
