@@ -212,25 +212,28 @@ void conv_wgrad_kernel(const __bf16* __restrict__ dy,
                        const __bf16* __restrict__ x,
                        float* __restrict__ dw, ConvDims d) {
   // d: M = Kout, N = R*S*C, K = Nb*P*Q; OH/OW = P,Q; GH/GW/GC = H,W,C
-  __shared__ __bf16 sA[64 * LDK];  // [kout][npq]
-  __shared__ __bf16 sB[64 * LDK];  // [rsc][npq]
+  // 64(Kout) x 128(rsc) tile: each loaded byte feeds twice the MFMA work of
+  // the 64x64 tile, and the split-K atomics halve. FAST needs C % 64 == 0 so
+  // every 32-wide rsc sub-chunk stays inside one (r,s) filter tap.
+  __shared__ __bf16 sA[64 * LDK];   // [kout][npq]
+  __shared__ __bf16 sB[128 * LDK];  // [rsc][npq]
 
   const int tid = threadIdx.x;
   const int m0 = blockIdx.y * 64;
-  const int n0 = blockIdx.x * 64;
-  const int row = tid & 63;       // npq row within the chunk
-  const int grp = (tid >> 6) * 16;  // 16-element column group
+  const int n0 = blockIdx.x * 128;
+  const int row = tid & 63;        // npq row within the chunk
+  const int grp = tid >> 6;        // 0..3
 
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int wm = (wave >> 1) * 32;
-  const int wn = (wave & 1) * 32;
+  const int wn = (wave & 1) * 64;
   const int fr = lane & 15;
   const int fk = (lane >> 4) * 8;
 
   const int SC = d.S * d.GC;
   const int nchunks = (d.K + BK - 1) / BK;
-  f32x4 acc[2][2] = {};
+  f32x4 acc[2][4] = {};
 
   for (int kc = blockIdx.z; kc < nchunks; kc += gridDim.z) {
     const int kk0 = kc * BK;
@@ -241,29 +244,32 @@ void conv_wgrad_kernel(const __bf16* __restrict__ dy,
 
     // ---- dy tile, transposed into sA[kout][npq] ------------------------
     {
+      const int cg = grp * 16;
       __bf16 vals[16];
-      if (npq_ok && (FAST || m0 + grp + 16 <= d.M)) {
-        const float4* src = (const float4*)(dy + (long)npq * d.M + m0 + grp);
+      if (npq_ok && (FAST || m0 + cg + 16 <= d.M)) {
+        const float4* src = (const float4*)(dy + (long)npq * d.M + m0 + cg);
         *(float4*)&vals[0] = src[0];
         *(float4*)&vals[8] = src[1];
       } else {
         #pragma unroll
         for (int e = 0; e < 16; ++e)
-          vals[e] = (npq_ok && m0 + grp + e < d.M)
-              ? dy[(long)npq * d.M + m0 + grp + e] : (__bf16)0.f;
+          vals[e] = (npq_ok && m0 + cg + e < d.M)
+              ? dy[(long)npq * d.M + m0 + cg + e] : (__bf16)0.f;
       }
       #pragma unroll
       for (int e = 0; e < 16; ++e)
-        sA[(grp + e) * LDK + row] = vals[e];
+        sA[(cg + e) * LDK + row] = vals[e];
     }
-    // ---- x tile, transposed into sB[rsc][npq] --------------------------
+    // ---- x tile (128 rsc cols = 4 x 32-col sub-chunks), transposed -----
     {
-      __bf16 vals[16];
+      const int cg = grp * 32;     // this thread's 32-col sub-chunk
+      __bf16 vals[32];
       if (FAST) {
-        const int i = n0 / SC;
-        const int rem = n0 - i * SC;
+        const int nn = n0 + cg;
+        const int i = nn / SC;
+        const int rem = nn - i * SC;
         const int j = rem / d.GC;
-        const int c0 = rem - j * d.GC + grp;
+        const int c0 = rem - j * d.GC;
         const int ih = xp * d.stride + i - d.pad;
         const int iw = xq * d.stride + j - d.pad;
         if (npq_ok && ih >= 0 && ih < d.GH && iw >= 0 && iw < d.GW) {
@@ -271,14 +277,16 @@ void conv_wgrad_kernel(const __bf16* __restrict__ dy,
               (((long)xn * d.GH + ih) * d.GW + iw) * d.GC + c0);
           *(float4*)&vals[0] = src[0];
           *(float4*)&vals[8] = src[1];
+          *(float4*)&vals[16] = src[2];
+          *(float4*)&vals[24] = src[3];
         } else {
           #pragma unroll
-          for (int e = 0; e < 16; ++e) vals[e] = (__bf16)0.f;
+          for (int e = 0; e < 32; ++e) vals[e] = (__bf16)0.f;
         }
       } else {
         #pragma unroll
-        for (int e = 0; e < 16; ++e) {
-          const int nn = n0 + grp + e;
+        for (int e = 0; e < 32; ++e) {
+          const int nn = n0 + cg + e;
           __bf16 v = (__bf16)0.f;
           if (npq_ok && nn < d.N) {
             const int i = nn / SC, rem = nn - i * SC;
@@ -292,24 +300,24 @@ void conv_wgrad_kernel(const __bf16* __restrict__ dy,
         }
       }
       #pragma unroll
-      for (int e = 0; e < 16; ++e)
-        sB[(grp + e) * LDK + row] = vals[e];
+      for (int e = 0; e < 32; ++e)
+        sB[(cg + e) * LDK + row] = vals[e];
     }
     __syncthreads();
 
     #pragma unroll
     for (int ks = 0; ks < BK; ks += 32) {
-      bf16x8 af[2], bf[2];
+      bf16x8 af[2], bf[4];
       #pragma unroll
       for (int mi = 0; mi < 2; ++mi)
         af[mi] = *(const bf16x8*)&sA[(wm + mi * 16 + fr) * LDK + ks + fk];
       #pragma unroll
-      for (int ni = 0; ni < 2; ++ni)
+      for (int ni = 0; ni < 4; ++ni)
         bf[ni] = *(const bf16x8*)&sB[(wn + ni * 16 + fr) * LDK + ks + fk];
       #pragma unroll
       for (int mi = 0; mi < 2; ++mi)
         #pragma unroll
-        for (int ni = 0; ni < 2; ++ni)
+        for (int ni = 0; ni < 4; ++ni)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
     }
@@ -321,7 +329,7 @@ void conv_wgrad_kernel(const __bf16* __restrict__ dy,
   #pragma unroll
   for (int mi = 0; mi < 2; ++mi)
     #pragma unroll
-    for (int ni = 0; ni < 2; ++ni) {
+    for (int ni = 0; ni < 4; ++ni) {
       const int n = n0 + wn + ni * 16 + dn;
       if (n >= d.N) continue;
       #pragma unroll
@@ -434,9 +442,9 @@ at::Tensor conv_wgrad_igemm(at::Tensor dy, at::Tensor x, long R, long S,
   auto dw = at::zeros({K, (long)(R * S * C)}, x.options().dtype(at::kFloat));
   ConvDims d{Nb, P, Q, H, W, C, (int)R, (int)S, (int)stride, (int)pad,
              K, (int)(R * S * C), Nb * P * Q};
-  const int tm = (d.M + 63) / 64, tn = (d.N + 63) / 64;
+  const int tm = (d.M + 63) / 64, tn = (d.N + 127) / 128;
   const int nchunks = (d.K + BK - 1) / BK;
-  int splits = 768 / (tm * tn);
+  int splits = 640 / (tm * tn);
   splits = std::max(1, std::min(splits, nchunks));
   const dim3 grid(tn, tm, splits);
   const bool fast = (C % BK == 0) && (K % 16 == 0);
