@@ -75,21 +75,27 @@ __device__ __forceinline__ bool tap_coord(int oh, int i, int filt, int stride,
 // fwd / dgrad kernel: out[M][N] = gatherA[M][K] @ B[N][K]^T
 // 256 threads = 4 waves as a 2x2 wave grid; per wave 64x32 via 4x2 MFMA tiles.
 // ---------------------------------------------------------------------------
-template <int MODE, bool FAST>
+// BMT = GEMM-M tile (128 default; 64 doubles the block count for the late
+// small-M stages so they still fill 256 CUs). MI = BMT/32 MFMA row-tiles per
+// wave; EPT = BMT/4 = A-tile elements each thread stages per BK chunk.
+template <int MODE, bool FAST, int BMT = BM>
 __global__ __launch_bounds__(256)
 void conv_igemm_kernel(const __bf16* __restrict__ Ag,
                        const __bf16* __restrict__ Bg,
                        __bf16* __restrict__ out, ConvDims d) {
-  __shared__ __bf16 sA[BM * LDK];
+  constexpr int MI = BMT / 32;
+  constexpr int EPT = BMT / 4;          // 32 or 16 elems per loader thread
+  constexpr int PER_ROW = 64 / EPT;     // loader threads per A row
+  __shared__ __bf16 sA[BMT * LDK];
   __shared__ __bf16 sB[BN * LDK];
 
   const int tid = threadIdx.x;
-  const int m0 = blockIdx.y * BM;
+  const int m0 = blockIdx.y * BMT;
   const int n0 = blockIdx.x * BN;
 
-  // A loader: thread -> (row, 32-element half of the BK chunk)
-  const int a_row = tid >> 1;
-  const int a_off = (tid & 1) * 32;
+  // A loader: thread -> (row, EPT-element slice of the BK chunk)
+  const int a_row = tid / PER_ROW;
+  const int a_off = (tid % PER_ROW) * EPT;
   int a_n, a_oh, a_ow;
   {
     int m = m0 + a_row;
@@ -103,12 +109,12 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
 
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int wm = (wave >> 1) * 64;
+  const int wm = (wave >> 1) * (BMT / 2);
   const int wn = (wave & 1) * 32;
   const int fr = lane & 15;
   const int fk = (lane >> 4) * 8;
 
-  f32x4 acc[4][2] = {};
+  f32x4 acc[MI][2] = {};
 
   const int SC = d.S * d.GC;
   for (int kk0 = 0; kk0 < d.K; kk0 += BK) {
@@ -121,17 +127,19 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
       int ih, iw;
       const bool ok = tap_coord<MODE>(a_oh, i, d.R, d.stride, d.pad, d.GH, ih)
                     & tap_coord<MODE>(a_ow, j, d.S, d.stride, d.pad, d.GW, iw);
-      float4 v0 = {}, v1 = {}, v2 = {}, v3 = {};
+      float4 v[EPT / 8] = {};
       if (ok) {
         const float4* src = (const float4*)(Ag +
             (((long)a_n * d.GH + ih) * d.GW + iw) * d.GC + c0);
-        v0 = src[0]; v1 = src[1]; v2 = src[2]; v3 = src[3];
+        #pragma unroll
+        for (int q = 0; q < EPT / 8; ++q) v[q] = src[q];
       }
       float4* dst = (float4*)(sA + a_row * LDK + a_off);
-      dst[0] = v0; dst[1] = v1; dst[2] = v2; dst[3] = v3;
+      #pragma unroll
+      for (int q = 0; q < EPT / 8; ++q) dst[q] = v[q];
     } else {
       // generic gather: one element at a time (stem conv only)
-      for (int e = tid; e < BM * BK; e += 256) {
+      for (int e = tid; e < BMT * BK; e += 256) {
         const int row = e >> 6, kk = kk0 + (e & 63);
         __bf16 v = (__bf16)0.f;
         if (kk < d.K) {
@@ -165,15 +173,15 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
 
     #pragma unroll
     for (int ks = 0; ks < BK; ks += 32) {
-      bf16x8 af[4], bf[2];
+      bf16x8 af[MI], bf[2];
       #pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
+      for (int mi = 0; mi < MI; ++mi)
         af[mi] = *(const bf16x8*)&sA[(wm + mi * 16 + fr) * LDK + ks + fk];
       #pragma unroll
       for (int ni = 0; ni < 2; ++ni)
         bf[ni] = *(const bf16x8*)&sB[(wn + ni * 16 + fr) * LDK + ks + fk];
       #pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
+      for (int mi = 0; mi < MI; ++mi)
         #pragma unroll
         for (int ni = 0; ni < 2; ++ni)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -186,7 +194,7 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
   const int dm = (lane >> 4) * 4;
   const int dn = lane & 15;
   #pragma unroll
-  for (int mi = 0; mi < 4; ++mi) {
+  for (int mi = 0; mi < MI; ++mi) {
     #pragma unroll
     for (int ni = 0; ni < 2; ++ni) {
       const int n = n0 + wn + ni * 16 + dn;
@@ -401,12 +409,22 @@ at::Tensor conv_fwd_igemm(at::Tensor x, at::Tensor w, long stride, long pad) {
                        x.options().memory_format(at::MemoryFormat::ChannelsLast));
   ConvDims d{Nb, P, Q, H, W, C, R, S, (int)stride, (int)pad,
              Nb * P * Q, K, R * S * C};
-  const dim3 grid((d.N + BN - 1) / BN, (d.M + BM - 1) / BM);
   const bool fast = (C % BK == 0);
-  auto* kern = fast ? conv_igemm_kernel<0, true> : conv_igemm_kernel<0, false>;
-  hipLaunchKernelGGL(kern, grid, dim3(256), 0, conv_stream(),
-                     bf16_ptr(x), bf16_ptr(w),
-                     reinterpret_cast<__bf16*>(out.data_ptr()), d);
+  const long tiles128 = (long)((d.M + BM - 1) / BM) * ((d.N + BN - 1) / BN);
+  if (tiles128 < 384) {  // small-M late stages: halve the tile, double blocks
+    const dim3 grid((d.N + BN - 1) / BN, (d.M + 63) / 64);
+    auto* kern = fast ? conv_igemm_kernel<0, true, 64>
+                      : conv_igemm_kernel<0, false, 64>;
+    hipLaunchKernelGGL(kern, grid, dim3(256), 0, conv_stream(),
+                       bf16_ptr(x), bf16_ptr(w),
+                       reinterpret_cast<__bf16*>(out.data_ptr()), d);
+  } else {
+    const dim3 grid((d.N + BN - 1) / BN, (d.M + BM - 1) / BM);
+    auto* kern = fast ? conv_igemm_kernel<0, true> : conv_igemm_kernel<0, false>;
+    hipLaunchKernelGGL(kern, grid, dim3(256), 0, conv_stream(),
+                       bf16_ptr(x), bf16_ptr(w),
+                       reinterpret_cast<__bf16*>(out.data_ptr()), d);
+  }
   return out;
 }
 
@@ -424,12 +442,22 @@ at::Tensor conv_dgrad_igemm(at::Tensor dy, at::Tensor wT, long H, long W,
                       dy.options().memory_format(at::MemoryFormat::ChannelsLast));
   ConvDims d{Nb, (int)H, (int)W, P, Q, K, R, S, (int)stride, (int)pad,
              Nb * (int)H * (int)W, C, R * S * K};
-  const dim3 grid((d.N + BN - 1) / BN, (d.M + BM - 1) / BM);
   const bool fast = (K % BK == 0);
-  auto* kern = fast ? conv_igemm_kernel<1, true> : conv_igemm_kernel<1, false>;
-  hipLaunchKernelGGL(kern, grid, dim3(256), 0, conv_stream(),
-                     bf16_ptr(dy), bf16_ptr(wT),
-                     reinterpret_cast<__bf16*>(dx.data_ptr()), d);
+  const long tiles128 = (long)((d.M + BM - 1) / BM) * ((d.N + BN - 1) / BN);
+  if (tiles128 < 384) {
+    const dim3 grid((d.N + BN - 1) / BN, (d.M + 63) / 64);
+    auto* kern = fast ? conv_igemm_kernel<1, true, 64>
+                      : conv_igemm_kernel<1, false, 64>;
+    hipLaunchKernelGGL(kern, grid, dim3(256), 0, conv_stream(),
+                       bf16_ptr(dy), bf16_ptr(wT),
+                       reinterpret_cast<__bf16*>(dx.data_ptr()), d);
+  } else {
+    const dim3 grid((d.N + BN - 1) / BN, (d.M + BM - 1) / BM);
+    auto* kern = fast ? conv_igemm_kernel<1, true> : conv_igemm_kernel<1, false>;
+    hipLaunchKernelGGL(kern, grid, dim3(256), 0, conv_stream(),
+                       bf16_ptr(dy), bf16_ptr(wT),
+                       reinterpret_cast<__bf16*>(dx.data_ptr()), d);
+  }
   return dx;
 }
 
