@@ -38,5 +38,45 @@ def kernel_stats(path, top=60):
     return "\n".join(out)
 
 
+
+
+def pmc_stats(path, top=30):
+    """Per-kernel PMC aggregates from a rocprofv3 --pmc run: MFMA-busy and
+    wait fractions of wave cycles."""
+    import collections
+    db = sqlite3.connect(path)
+    tabs = [r[0] for r in db.execute(
+        "SELECT name FROM sqlite_master WHERE type='table'")]
+    suffix = [t for t in tabs if t.startswith("rocpd_kernel_dispatch")][0][
+        len("rocpd_kernel_dispatch"):]
+    pmc_names = dict(db.execute(f"SELECT id, name FROM rocpd_info_pmc{suffix}"))
+    q = f"""
+      SELECT ks.display_name, p.pmc_id, SUM(p.value), COUNT(*)
+      FROM rocpd_pmc_event{suffix} p
+      JOIN rocpd_kernel_dispatch{suffix} k ON k.event_id = p.event_id
+      JOIN rocpd_info_kernel_symbol{suffix} ks ON ks.id = k.kernel_id
+      GROUP BY ks.display_name, p.pmc_id"""
+    agg = collections.defaultdict(dict)
+    for name, pid, val, cnt in db.execute(q):
+        agg[name][pmc_names[pid]] = val
+    rows = []
+    for name, d in agg.items():
+        wc = d.get("SQ_WAVE_CYCLES", 0) or 1
+        rows.append((name, d.get("SQ_VALU_MFMA_BUSY_CYCLES", 0),
+                     d.get("SQ_WAIT_ANY", 0), wc))
+    rows.sort(key=lambda r: -r[3])
+    out = [f"{'mfma%':>7} {'wait%':>7} {'wave_Mcyc':>10}  kernel",
+           "(mfma% = SQ_VALU_MFMA_BUSY_CYCLES/quad / SQ_WAVE_CYCLES;"
+           " wait% = SQ_WAIT_ANY / SQ_WAVE_CYCLES)"]
+    for name, mfma, wait, wc in rows[:top]:
+        out.append(f"{100*(mfma/4)/wc:7.2f} {100*wait/wc:7.2f} "
+                   f"{wc/1e6:10.1f}  {name[:120]}")
+    return "\n".join(out)
+
+
 if __name__ == "__main__":
-    print(kernel_stats(sys.argv[1], int(sys.argv[2]) if len(sys.argv) > 2 else 60))
+    if len(sys.argv) > 2 and sys.argv[2] == "pmc":
+        print(pmc_stats(sys.argv[1]))
+    else:
+        print(kernel_stats(sys.argv[1],
+                           int(sys.argv[2]) if len(sys.argv) > 2 else 60))
