@@ -56,8 +56,11 @@ __device__ __forceinline__ void decode_m(int m, const ConvDims& d,
 }
 
 // Input row coordinate for filter tap i. MODE 0 = fwd (oh is an output pixel,
-// gather from x), MODE 1 = dgrad (oh is an input pixel, gather from dy with
+// gather from x); MODE 1/2 = dgrad (oh is an input pixel, gather from dy with
 // the transpose-conv index relation; tap index i is the 180°-rotated r).
+// MODE 2 is the stride-2 parity-decomposed dgrad: each block handles one
+// (h%2, w%2) output class, so filter taps whose divisibility test fails for
+// the whole class are skipped wholesale (see the k-loop).
 template <int MODE>
 __device__ __forceinline__ bool tap_coord(int oh, int i, int filt, int stride,
                                           int pad, int lim, int& ih) {
@@ -69,6 +72,17 @@ __device__ __forceinline__ bool tap_coord(int oh, int i, int filt, int stride,
   if (t < 0 || t % stride != 0) return false;
   ih = t / stride;
   return ih < lim;
+}
+
+// MODE 2 row decode: class-local m -> real (n, oh, ow) for parity (pa, pb).
+__device__ __forceinline__ void decode_m2(int m, const ConvDims& d,
+                                          int pa, int pb,
+                                          int& n, int& oh, int& ow) {
+  const int ow2 = d.OW >> 1, oh2 = d.OH >> 1;
+  ow = (m % ow2) * 2 + pb;
+  const int t = m / ow2;
+  oh = (t % oh2) * 2 + pa;
+  n = t / oh2;
 }
 
 // ---------------------------------------------------------------------------
@@ -92,6 +106,8 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
   const int tid = threadIdx.x;
   const int m0 = blockIdx.y * BMT;
   const int n0 = blockIdx.x * BN;
+  const int pa = (MODE == 2) ? (int)(blockIdx.z & 1) : 0;
+  const int pb = (MODE == 2) ? (int)(blockIdx.z >> 1) : 0;
 
   // A loader: thread -> (row, EPT-element slice of the BK chunk)
   const int a_row = tid / PER_ROW;
@@ -100,7 +116,8 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
   {
     int m = m0 + a_row;
     if (m >= d.M) m = d.M - 1;  // clamped rows only feed predicated-out outputs
-    decode_m(m, d, a_n, a_oh, a_ow);
+    if (MODE == 2) decode_m2(m, d, pa, pb, a_n, a_oh, a_ow);
+    else decode_m(m, d, a_n, a_oh, a_ow);
   }
   // B loader: thread -> (row, 16-element quarter of the BK chunk)
   const int b_row = tid & 63;
@@ -124,6 +141,13 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
       const int rem = kk0 - i * SC;
       const int j = rem / d.GC;
       const int c0 = rem - j * d.GC + a_off;
+      if (MODE == 2) {
+        // stride-2 divisibility is class-uniform: skip the whole tap when
+        // either axis has the wrong parity (3/4 of taps for a 3x3 s2 dgrad)
+        const int ta = pa + d.pad - d.R + 1 + i;
+        const int tb = pb + d.pad - d.S + 1 + j;
+        if ((ta & 1) || (tb & 1)) continue;
+      }
       int ih, iw;
       const bool ok = tap_coord<MODE>(a_oh, i, d.R, d.stride, d.pad, d.GH, ih)
                     & tap_coord<MODE>(a_ow, j, d.S, d.stride, d.pad, d.GW, iw);
@@ -212,7 +236,14 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
       #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int m = m0 + wm + mi * 16 + dm + r;
-        if (m < d.M) out[(long)m * d.N + n] = (__bf16)acc[mi][ni][r];
+        if (m >= d.M) continue;
+        long om = m;
+        if (MODE == 2) {
+          int on, ooh, oow;
+          decode_m2(m, d, pa, pb, on, ooh, oow);
+          om = ((long)on * d.OH + ooh) * d.OW + oow;
+        }
+        out[om * d.N + n] = (__bf16)acc[mi][ni][r];
       }
     }
   }
@@ -453,6 +484,25 @@ at::Tensor conv_dgrad_igemm(at::Tensor dy, at::Tensor wT, long H, long W,
   ConvDims d{Nb, (int)H, (int)W, P, Q, K, R, S, (int)stride, (int)pad,
              Nb * (int)H * (int)W, C, R * S * K};
   const bool fast = (K % BK == 0);
+  if (fast && stride == 2 && H % 2 == 0 && W % 2 == 0) {
+    // parity-decomposed: one grid.z slice per (h%2, w%2) output class,
+    // class-local M; wrong-parity taps are skipped inside the kernel.
+    ConvDims d2 = d;
+    d2.M = Nb * (int)(H / 2) * (int)(W / 2);
+    const long tiles = (long)((d2.M + BM - 1) / BM) * ((d2.N + BN - 1) / BN) * 4;
+    if (tiles < 384) {
+      const dim3 grid((d2.N + BN - 1) / BN, (d2.M + 63) / 64, 4);
+      hipLaunchKernelGGL((conv_igemm_kernel<2, true, 64>), grid, dim3(256), 0,
+                         conv_stream(), bf16_ptr(dy), bf16_ptr(wT),
+                         reinterpret_cast<__bf16*>(dx.data_ptr()), d2);
+    } else {
+      const dim3 grid((d2.N + BN - 1) / BN, (d2.M + BM - 1) / BM, 4);
+      hipLaunchKernelGGL((conv_igemm_kernel<2, true, 128>), grid, dim3(256), 0,
+                         conv_stream(), bf16_ptr(dy), bf16_ptr(wT),
+                         reinterpret_cast<__bf16*>(dx.data_ptr()), d2);
+    }
+    return dx;
+  }
   const long tiles128 = (long)((d.M + BM - 1) / BM) * ((d.N + BN - 1) / BN);
   if (tiles128 < 384) {
     const dim3 grid((d.N + BN - 1) / BN, (d.M + 63) / 64);

@@ -113,14 +113,15 @@ __global__ void bn_fwd_kernel(const scalar_t* __restrict__ x,
 // Vectorized variant: 8 elements (16 B) per thread. Correct when the
 // channel run stays within one c-block: CLAST needs C %% 8 == 0, NCHW needs
 // S %% 8 == 0 (true for every ResNet CIFAR stage).
+// y = x*scale[c] + shift[c] (+res, relu): the four per-channel tables
+// (mean/invstd/weight/bias) are pre-folded into two by bn_coeffs, and in
+// CLAST mode the 8 consecutive channels load as two float4s.
 template <typename scalar_t, bool RELU, bool HAS_RES, bool CLAST>
 __global__ void bn_fwd_vec_kernel(const scalar_t* __restrict__ x,
                                   const scalar_t* __restrict__ res,
                                   scalar_t* __restrict__ y,
-                                  const float* __restrict__ weight,
-                                  const float* __restrict__ bias,
-                                  const float* __restrict__ mean,
-                                  const float* __restrict__ invstd,
+                                  const float* __restrict__ scale,
+                                  const float* __restrict__ shift,
                                   long nvec, int C, int S) {
   for (long v = (long)blockIdx.x * blockDim.x + threadIdx.x; v < nvec;
        v += (long)gridDim.x * blockDim.x) {
@@ -128,17 +129,67 @@ __global__ void bn_fwd_vec_kernel(const scalar_t* __restrict__ x,
     scalar_t x8[8], r8[8], y8[8];
     *(float4*)x8 = *(const float4*)(x + i);
     if (HAS_RES) *(float4*)r8 = *(const float4*)(res + i);
-    const int cbase = CLAST ? (int)(i % C) : (int)((i / S) % C);
+    float sc[8], sh[8];
+    if (CLAST) {
+      const int c0 = (int)(i % C);
+      *(float4*)&sc[0] = *(const float4*)(scale + c0);
+      *(float4*)&sc[4] = *(const float4*)(scale + c0 + 4);
+      *(float4*)&sh[0] = *(const float4*)(shift + c0);
+      *(float4*)&sh[4] = *(const float4*)(shift + c0 + 4);
+    } else {
+      const int c = (int)((i / S) % C);
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) { sc[e] = scale[c]; sh[e] = shift[c]; }
+    }
     #pragma unroll
     for (int e = 0; e < 8; ++e) {
-      const int c = CLAST ? cbase + e : cbase;
-      float o = ((float)x8[e] - mean[c]) * invstd[c] * weight[c] + bias[c];
+      float o = (float)x8[e] * sc[e] + sh[e];
       if (HAS_RES) o += (float)r8[e];
       if (RELU) o = fmaxf(o, 0.f);
       y8[e] = (scalar_t)o;
     }
     *(float4*)(y + i) = *(float4*)y8;
   }
+}
+
+// scale = w*invstd, shift = b - mean*scale (one tiny launch per BN forward)
+__global__ void bn_coeffs_kernel(const float* __restrict__ weight,
+                                 const float* __restrict__ bias,
+                                 const float* __restrict__ mean,
+                                 const float* __restrict__ invstd,
+                                 float* __restrict__ scale,
+                                 float* __restrict__ shift, int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float sc = weight[c] * invstd[c];
+  scale[c] = sc;
+  shift[c] = bias[c] - mean[c] * sc;
+}
+
+// backward coefficient fold: dx = A[c]*g + B[c]*x + D[c]
+//   A = w*is; B = -w*is^2*sum_dy_xhat/cnt; D = -A*sum_dy/cnt - B*mean
+// (eval mode: B = D = 0 -> dx = A*g)
+__global__ void bn_bwd_coeffs_kernel(const float* __restrict__ weight,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ invstd,
+                                     const float* __restrict__ sum_dy,
+                                     const float* __restrict__ sum_dy_xhat,
+                                     float inv_count, bool training,
+                                     float* __restrict__ A,
+                                     float* __restrict__ B,
+                                     float* __restrict__ D, int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float is = invstd[c];
+  const float a = weight[c] * is;
+  float b = 0.f, dd = 0.f;
+  if (training) {
+    b = -a * is * sum_dy_xhat[c] * inv_count;
+    dd = -a * sum_dy[c] * inv_count - b * mean[c];
+  }
+  A[c] = a;
+  B[c] = b;
+  D[c] = dd;
 }
 
 // ---------------------------------------------------------------------------
@@ -212,18 +263,18 @@ __global__ void bn_bwd_kernel(const scalar_t* __restrict__ dy,
 }
 
 
+// dx = A[c]*g + B[c]*x + D[c] with the relu mask on g; coefficient tables
+// from bn_bwd_coeffs_kernel (eval: B=D=0).
 template <typename scalar_t, bool RELU, bool TRAIN, bool NEED_DRES, bool CLAST>
 __global__ void bn_bwd_vec_kernel(const scalar_t* __restrict__ dy,
                                   const scalar_t* __restrict__ x,
                                   const scalar_t* __restrict__ y,
                                   scalar_t* __restrict__ dx,
                                   scalar_t* __restrict__ dres,
-                                  const float* __restrict__ weight,
-                                  const float* __restrict__ mean,
-                                  const float* __restrict__ invstd,
-                                  const float* __restrict__ sum_dy,
-                                  const float* __restrict__ sum_dy_xhat,
-                                  float inv_count, long nvec, int C, int S) {
+                                  const float* __restrict__ A,
+                                  const float* __restrict__ B,
+                                  const float* __restrict__ D,
+                                  long nvec, int C, int S) {
   for (long v = (long)blockIdx.x * blockDim.x + threadIdx.x; v < nvec;
        v += (long)gridDim.x * blockDim.x) {
     const long i = v * 8;
@@ -231,21 +282,32 @@ __global__ void bn_bwd_vec_kernel(const scalar_t* __restrict__ dy,
     *(float4*)dy8 = *(const float4*)(dy + i);
     if (TRAIN) *(float4*)x8 = *(const float4*)(x + i);
     if (RELU) *(float4*)y8 = *(const float4*)(y + i);
-    const int cbase = CLAST ? (int)(i % C) : (int)((i / S) % C);
+    float a[8], b[8], dd[8];
+    if (CLAST) {
+      const int c0 = (int)(i % C);
+      *(float4*)&a[0] = *(const float4*)(A + c0);
+      *(float4*)&a[4] = *(const float4*)(A + c0 + 4);
+      if (TRAIN) {
+        *(float4*)&b[0] = *(const float4*)(B + c0);
+        *(float4*)&b[4] = *(const float4*)(B + c0 + 4);
+        *(float4*)&dd[0] = *(const float4*)(D + c0);
+        *(float4*)&dd[4] = *(const float4*)(D + c0 + 4);
+      }
+    } else {
+      const int c = (int)((i / S) % C);
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        a[e] = A[c];
+        if (TRAIN) { b[e] = B[c]; dd[e] = D[c]; }
+      }
+    }
     #pragma unroll
     for (int e = 0; e < 8; ++e) {
-      const int c = CLAST ? cbase + e : cbase;
       float g = (float)dy8[e];
       if (RELU && (float)y8[e] <= 0.f) g = 0.f;
       if (NEED_DRES) dr8[e] = (scalar_t)g;
-      const float w_is = weight[c] * invstd[c];
-      float o;
-      if (TRAIN) {
-        const float xhat = ((float)x8[e] - mean[c]) * invstd[c];
-        o = w_is * (g - sum_dy[c] * inv_count - xhat * sum_dy_xhat[c] * inv_count);
-      } else {
-        o = w_is * g;
-      }
+      float o = a[e] * g;
+      if (TRAIN) o += b[e] * (float)x8[e] + dd[e];
       o8[e] = (scalar_t)o;
     }
     *(float4*)(dx + i) = *(float4*)o8;
@@ -899,6 +961,13 @@ at::Tensor bn_fwd(at::Tensor x, at::Tensor weight, at::Tensor bias,
       x.scalar_type(), "bn_fwd", [&] {
     auto launch = [&](auto relu_c, auto res_c, auto cl_c) {
       if (vec_ok && sizeof(scalar_t) == 2) {
+        auto coeffs = at::empty({2, C}, x.options().dtype(at::kFloat));
+        float* sc = coeffs.data_ptr<float>();
+        float* sh = sc + C;
+        hipLaunchKernelGGL(bn_coeffs_kernel, dim3((C + 255) / 256), dim3(256),
+                           0, cur_stream(), wf.data_ptr<float>(),
+                           bf.data_ptr<float>(), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), sc, sh, C);
         const long nvec = total / 8;
         const int vblocks = (int)std::min((nvec + 255) / 256, (long)2048);
         hipLaunchKernelGGL((bn_fwd_vec_kernel<scalar_t, decltype(relu_c)::value,
@@ -907,9 +976,7 @@ at::Tensor bn_fwd(at::Tensor x, at::Tensor weight, at::Tensor bias,
                            dim3(vblocks), dim3(256), 0, cur_stream(),
                            x.data_ptr<scalar_t>(),
                            has_res ? residual.data_ptr<scalar_t>() : nullptr,
-                           y.data_ptr<scalar_t>(), wf.data_ptr<float>(),
-                           bf.data_ptr<float>(), mean.data_ptr<float>(),
-                           invstd.data_ptr<float>(), nvec, C, S);
+                           y.data_ptr<scalar_t>(), sc, sh, nvec, C, S);
         return;
       }
       hipLaunchKernelGGL((bn_fwd_kernel<scalar_t, decltype(relu_c)::value,
@@ -997,6 +1064,16 @@ std::vector<at::Tensor> bn_bwd(at::Tensor dy, at::Tensor x, at::Tensor weight,
         (clast ? (C % 8 == 0) : (S % 8 == 0));
     auto launch = [&](auto relu_c, auto train_c, auto dres_c, auto cl_c) {
       if (vec_ok && sizeof(scalar_t) == 2) {
+        auto coeffs = at::empty({3, C}, x.options().dtype(at::kFloat));
+        float* A = coeffs.data_ptr<float>();
+        float* B = A + C;
+        float* D = B + C;
+        hipLaunchKernelGGL(bn_bwd_coeffs_kernel, dim3((C + 255) / 256),
+                           dim3(256), 0, cur_stream(), wf.data_ptr<float>(),
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           sum_dy.data_ptr<float>(),
+                           sum_dy_xhat.data_ptr<float>(),
+                           (float)(1.0 / count), training, A, B, D, C);
         const long nvec = total / 8;
         const int vblocks = (int)std::min((nvec + 255) / 256, (long)2048);
         hipLaunchKernelGGL((bn_bwd_vec_kernel<scalar_t, decltype(relu_c)::value,
@@ -1007,10 +1084,7 @@ std::vector<at::Tensor> bn_bwd(at::Tensor dy, at::Tensor x, at::Tensor weight,
                            dy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
                            y.data_ptr<scalar_t>(), dx.data_ptr<scalar_t>(),
                            need_dres ? dres.data_ptr<scalar_t>() : nullptr,
-                           wf.data_ptr<float>(), mean.data_ptr<float>(),
-                           invstd.data_ptr<float>(), sum_dy.data_ptr<float>(),
-                           sum_dy_xhat.data_ptr<float>(),
-                           (float)(1.0 / count), nvec, C, S);
+                           A, B, D, nvec, C, S);
         return;
       }
       hipLaunchKernelGGL((bn_bwd_kernel<scalar_t, decltype(relu_c)::value,
