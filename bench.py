@@ -36,7 +36,10 @@ def parse_args():
     p.add_argument("--mode", type=str, default="flat", choices=["flat", "torchddp"])
     p.add_argument("--grad-accu-steps", type=int, default=1)
     p.add_argument("--no-syncbn", action="store_true")
-    p.add_argument("--channels-last", action="store_true")
+    p.add_argument("--channels-last", action="store_true", default=True,
+                   help="NHWC + native implicit-GEMM conv kernels (default)")
+    p.add_argument("--no-channels-last", dest="channels_last",
+                   action="store_false")
     p.add_argument("--hip-graph", action="store_true")
     p.add_argument("--image-size", type=int, default=32)
     return p.parse_args()
