@@ -178,14 +178,6 @@ def _group_for(bn: nn.Module):
     return None
 
 
-def _exponential_momentum(bn: nn.Module) -> float:
-    # nn.BatchNorm semantics: momentum=None means cumulative moving average.
-    if bn.momentum is None:
-        bn.num_batches_tracked += 0  # handled below in bn_relu
-        return 0.1
-    return bn.momentum
-
-
 def _run_fused(x, residual, bn: nn.Module, relu: bool):
     training = bn.training or not bn.track_running_stats
     if bn.training and bn.track_running_stats and bn.num_batches_tracked is not None:
