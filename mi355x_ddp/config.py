@@ -65,6 +65,9 @@ class TrainConfig:
     save_epoch: int = 15
     eval_every_epoch: bool = True
     resume: str = ""
+    # step caps (None = full epoch); used by smoke tests and quick CLI runs
+    max_train_steps: Optional[int] = None
+    max_eval_steps: Optional[int] = None
 
     def per_rank_batch(self, world_size: int) -> int:
         return max(1, self.batch_size // max(1, world_size))
@@ -95,6 +98,9 @@ def add_common_args(p: argparse.ArgumentParser) -> argparse.ArgumentParser:
     p.add_argument("--resume", default="", type=str)
     p.add_argument("--synthetic", action="store_true", default=None,
                    help="force synthetic CIFAR-shaped data (default: auto)")
+    p.add_argument("--max_train_steps", default=None, type=int,
+                   help="cap steps per epoch (smoke runs)")
+    p.add_argument("--max_eval_steps", default=None, type=int)
     return p
 
 
@@ -118,5 +124,9 @@ def config_from_args(args: argparse.Namespace, **overrides) -> TrainConfig:
         kw["synthetic"] = bool(args.synthetic)
     if getattr(args, "grad_accu_steps", None):
         kw["grad_accu_steps"] = args.grad_accu_steps
+    if getattr(args, "max_train_steps", None) is not None:
+        kw["max_train_steps"] = args.max_train_steps
+    if getattr(args, "max_eval_steps", None) is not None:
+        kw["max_eval_steps"] = args.max_eval_steps
     kw.update(overrides)
     return TrainConfig(**kw)

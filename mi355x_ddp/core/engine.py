@@ -122,6 +122,8 @@ def validate(model, loader, criterion, device, cfg: TrainConfig,
                              prefix="Test: ")
     end = time.time()
     for i, (images, labels) in enumerate(loader):
+        if cfg.max_eval_steps is not None and i >= cfg.max_eval_steps:
+            break
         images = images.to(device, non_blocking=True)
         labels = labels.to(device, non_blocking=True)
         if cfg.channels_last:
@@ -162,7 +164,8 @@ def fit(model, train_loader, test_loader, train_sampler, criterion, optimizer,
             train_sampler.set_epoch(epoch)
         t0 = time.time()
         train_one_epoch(model, train_loader, criterion, optimizer, epoch, cfg,
-                        device, scaler=scaler, sink=sink)
+                        device, scaler=scaler, sink=sink,
+                        max_steps=cfg.max_train_steps)
         scheduler.step()
         epoch_s = time.time() - t0
         if dist_utils.is_main_process():

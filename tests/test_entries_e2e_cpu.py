@@ -1,0 +1,60 @@
+"""End-to-end entry-script runs on CPU (gloo): each CLI trains 2 capped steps
+of one epoch on synthetic data, evaluates 2 batches, and writes a final
+checkpoint — the L5->L1 vertical slice of SURVEY.md §1 without a GPU."""
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+COMMON = ["--epochs", "1", "--batch_size", "64", "--num_workers", "0",
+          "--synthetic", "--max_train_steps", "2", "--max_eval_steps", "2",
+          "--log_interval", "1", "--save_epoch", "0", "--seed", "3"]
+
+
+def _run(tmp_path, script, extra, timeout=420):
+    env = dict(os.environ, PYTHONPATH=ROOT, MASTER_ADDR="127.0.0.1")
+    r = subprocess.run(
+        [sys.executable, os.path.join(ROOT, script)] + COMMON + extra,
+        cwd=tmp_path, env=env, capture_output=True, text=True, timeout=timeout)
+    assert r.returncode == 0, f"{script} failed:\n{r.stdout[-2000:]}\n{r.stderr[-2000:]}"
+    return r
+
+
+def test_distributed_single_process(tmp_path, free_port):
+    r = _run(tmp_path, "distributed.py", ["--port", str(free_port)])
+    assert "Epoch: [0][0/" in r.stdout
+    assert "Acc@1" in r.stdout
+    assert (tmp_path / "ckpts").exists()  # final checkpoint
+    metrics = tmp_path / "runs" / "metrics.jsonl"
+    assert metrics.exists()
+    kinds = {json.loads(l)["kind"] for l in metrics.read_text().splitlines()}
+    assert {"train", "epoch", "val"} <= kinds
+
+
+def test_distributed_mp_two_ranks(tmp_path, free_port):
+    r = _run(tmp_path, "distributed_mp.py",
+             ["--nprocs", "2", "--port", str(free_port)])
+    assert "Acc@1" in r.stdout
+
+
+def test_grad_accumulation_entry(tmp_path, free_port):
+    r = _run(tmp_path, "distributed_gradient_accumulation.py",
+             ["--grad_accu_steps", "2", "--port", str(free_port)])
+    assert "Acc@1" in r.stdout
+
+
+def test_dataparallel_entry(tmp_path):
+    r = _run(tmp_path, "dataparallel.py", [])
+    assert "Acc@1" in r.stdout
+
+
+def test_apex_equivalent_entry_fp16_scaler(tmp_path, free_port):
+    """distributed_apex.py on CPU: fp32 fallback for autocast-free CPU mode is
+    exercised via --amp fp32; the scaler path is covered by unit tests."""
+    r = _run(tmp_path, "distributed_apex.py",
+             ["--amp", "fp32", "--port", str(free_port)])
+    assert "Acc@1" in r.stdout
