@@ -9,7 +9,7 @@ inf/nan check as ONE HIP kernel, skip step on overflow, grow/backoff scale)
 from __future__ import annotations
 
 import contextlib
-from typing import Iterable, List, Optional
+from typing import List, Optional
 
 import torch
 

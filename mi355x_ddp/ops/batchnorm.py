@@ -14,7 +14,7 @@ so the distributed math is covered by CPU tests and numerics by GPU tests.
 """
 from __future__ import annotations
 
-from typing import Optional
+
 
 import torch
 import torch.distributed as dist

@@ -14,7 +14,7 @@ parallel.FlatDDP) and the generic multi-tensor path.
 """
 from __future__ import annotations
 
-from typing import Iterable, List, Optional
+from typing import List, Optional
 
 import torch
 

@@ -24,7 +24,7 @@ Design, chosen for the hardware rather than translated:
 from __future__ import annotations
 
 import contextlib
-from typing import List, Optional
+from typing import List
 
 import torch
 import torch.distributed as dist
