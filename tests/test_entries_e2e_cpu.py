@@ -58,3 +58,23 @@ def test_apex_equivalent_entry_fp16_scaler(tmp_path, free_port):
     r = _run(tmp_path, "distributed_apex.py",
              ["--amp", "fp32", "--port", str(free_port)])
     assert "Acc@1" in r.stdout
+
+
+def test_bench_multirank_contract(tmp_path, free_port):
+    """The driver's exact multi-rank launch: torchrun x2 on CPU (gloo),
+    one JSON line from rank 0 with the whole-job aggregate."""
+    env = dict(os.environ, PYTHONPATH=ROOT)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(free_port),
+         os.path.join(ROOT, "bench.py"), "--gpus", "2", "--steps", "2",
+         "--warmup", "1", "--global-batch", "32"],
+        cwd=tmp_path, env=env, capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stderr[-2000:]
+    lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, r.stdout
+    rec = json.loads(lines[0])
+    assert rec["n_gpus"] == 2 and rec["config"]["parallelism"] == "dp2"
+    assert rec["config"]["sync_bn"] is True
+    assert rec["value"] > 0 and rec["unit"] == "images/sec"
