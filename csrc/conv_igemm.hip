@@ -134,77 +134,49 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
   f32x4 acc[MI][2] = {};
 
   const int SC = d.S * d.GC;
-  for (int kk0 = 0; kk0 < d.K; kk0 += BK) {
-    // ---- stage A tile -------------------------------------------------
-    if (FAST) {
-      const int i = kk0 / SC;
-      const int rem = kk0 - i * SC;
-      const int j = rem / d.GC;
-      const int c0 = rem - j * d.GC + a_off;
-      if (MODE == 2) {
-        // stride-2 divisibility is class-uniform: skip the whole tap when
-        // either axis has the wrong parity (3/4 of taps for a 3x3 s2 dgrad)
-        const int ta = pa + d.pad - d.R + 1 + i;
-        const int tb = pb + d.pad - d.S + 1 + j;
-        if ((ta & 1) || (tb & 1)) continue;
-      }
-      int ih, iw;
-      const bool ok = tap_coord<MODE>(a_oh, i, d.R, d.stride, d.pad, d.GH, ih)
-                    & tap_coord<MODE>(a_ow, j, d.S, d.stride, d.pad, d.GW, iw);
-      // explicit scalars, not an array: an array here is demoted to scratch
-      // (80 B/lane spill measured) and costs 2-3x on the whole kernel
-      float4 v0 = {}, v1 = {}, v2 = {}, v3 = {};
-      if (ok) {
-        const float4* src = (const float4*)(Ag +
-            (((long)a_n * d.GH + ih) * d.GW + iw) * d.GC + c0);
-        v0 = src[0];
-        v1 = src[1];
-        if constexpr (EPT == 32) {
-          v2 = src[2];
-          v3 = src[3];
-        }
-      }
-      float4* dst = (float4*)(sA + a_row * LDK + a_off);
-      dst[0] = v0;
-      dst[1] = v1;
-      if constexpr (EPT == 32) {
-        dst[2] = v2;
-        dst[3] = v3;
-      }
-    } else {
-      // generic gather: one element at a time (stem conv only)
-      for (int e = tid; e < BMT * BK; e += 256) {
-        const int row = e >> 6, kk = kk0 + (e & 63);
-        __bf16 v = (__bf16)0.f;
-        if (kk < d.K) {
-          int m = m0 + row;
-          if (m >= d.M) m = d.M - 1;
-          int n, oh, ow;
-          decode_m(m, d, n, oh, ow);
-          const int i = kk / SC, rem = kk - i * SC;
-          const int j = rem / d.GC, c = rem - j * d.GC;
-          int ih, iw;
-          if (tap_coord<MODE>(oh, i, d.R, d.stride, d.pad, d.GH, ih) &&
-              tap_coord<MODE>(ow, j, d.S, d.stride, d.pad, d.GW, iw))
-            v = Ag[(((long)n * d.GH + ih) * d.GW + iw) * d.GC + c];
-        }
-        sA[row * LDK + (e & 63)] = v;
-      }
-    }
-    // ---- stage B tile (dense rows of w / wT) ---------------------------
-    if (FAST) {
-      const float4* src = (const float4*)(Bg + b_base + kk0 + b_off);
-      float4* dst = (float4*)(sB + b_row * LDK + b_off);
-      dst[0] = src[0]; dst[1] = src[1];
-    } else {
-      for (int e = tid; e < BN * BK; e += 256) {
-        const int row = e >> 6, kk = kk0 + (e & 63);
-        sB[row * LDK + (e & 63)] = (kk < d.K)
-            ? Bg[(long)min(n0 + row, d.N - 1) * d.K + kk] : (__bf16)0.f;
-      }
-    }
-    __syncthreads();
 
+  // ---- FAST-path staging helpers (explicit scalars: an array here is
+  // demoted to scratch — an 80 B/lane spill measured 2-3x on the kernel) ----
+  auto issue_A = [&](int kk0, float4& v0, float4& v1, float4& v2, float4& v3) {
+    const int i = kk0 / SC;
+    const int rem = kk0 - i * SC;
+    const int j = rem / d.GC;
+    const int c0 = rem - j * d.GC + a_off;
+    int ih, iw;
+    const bool ok = tap_coord<MODE>(a_oh, i, d.R, d.stride, d.pad, d.GH, ih)
+                  & tap_coord<MODE>(a_ow, j, d.S, d.stride, d.pad, d.GW, iw);
+    v0 = v1 = v2 = v3 = float4{};
+    if (ok) {
+      const float4* src = (const float4*)(Ag +
+          (((long)a_n * d.GH + ih) * d.GW + iw) * d.GC + c0);
+      v0 = src[0];
+      v1 = src[1];
+      if constexpr (EPT == 32) {
+        v2 = src[2];
+        v3 = src[3];
+      }
+    }
+  };
+  auto write_A = [&](float4 v0, float4 v1, float4 v2, float4 v3) {
+    float4* dst = (float4*)(sA + a_row * LDK + a_off);
+    dst[0] = v0;
+    dst[1] = v1;
+    if constexpr (EPT == 32) {
+      dst[2] = v2;
+      dst[3] = v3;
+    }
+  };
+  auto issue_B = [&](int kk0, float4& v0, float4& v1) {
+    const float4* src = (const float4*)(Bg + b_base + kk0 + b_off);
+    v0 = src[0];
+    v1 = src[1];
+  };
+  auto write_B = [&](float4 v0, float4 v1) {
+    float4* dst = (float4*)(sB + b_row * LDK + b_off);
+    dst[0] = v0;
+    dst[1] = v1;
+  };
+  auto mfma_tile = [&]() {
     #pragma unroll
     for (int ks = 0; ks < BK; ks += 32) {
       bf16x8 af[MI], bf[2];
@@ -221,7 +193,83 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
     }
+  };
+
+  if constexpr (FAST && MODE != 2) {
+    // Register-pipelined: tile k+1's global loads issue before tile k's
+    // MFMAs, so HBM/L2 latency overlaps compute; the waits land at the LDS
+    // write after the barrier (write-after-barrier form).
+    {
+      float4 a0, a1, a2, a3, b0, b1;
+      issue_A(0, a0, a1, a2, a3);
+      issue_B(0, b0, b1);
+      write_A(a0, a1, a2, a3);
+      write_B(b0, b1);
+    }
     __syncthreads();
+    for (int kk0 = 0; kk0 < d.K; kk0 += BK) {
+      const bool has_next = kk0 + BK < d.K;
+      float4 a0 = {}, a1 = {}, a2 = {}, a3 = {}, b0 = {}, b1 = {};
+      if (has_next) {
+        issue_A(kk0 + BK, a0, a1, a2, a3);
+        issue_B(kk0 + BK, b0, b1);
+      }
+      mfma_tile();
+      if (has_next) {
+        __syncthreads();
+        write_A(a0, a1, a2, a3);
+        write_B(b0, b1);
+        __syncthreads();
+      }
+    }
+  } else {
+    for (int kk0 = 0; kk0 < d.K; kk0 += BK) {
+      // ---- stage A tile ---------------------------------------------
+      if (FAST) {  // MODE == 2 only (other FAST modes take the pipeline)
+        if (MODE == 2) {
+          // stride-2 divisibility is class-uniform: skip the whole tap when
+          // either axis has the wrong parity (3/4 of taps for 3x3 s2 dgrad)
+          const int i = kk0 / SC;
+          const int rem = kk0 - i * SC;
+          const int j = rem / d.GC;
+          const int ta = pa + d.pad - d.R + 1 + i;
+          const int tb = pb + d.pad - d.S + 1 + j;
+          if ((ta & 1) || (tb & 1)) continue;
+        }
+        float4 a0, a1, a2, a3, b0, b1;
+        issue_A(kk0, a0, a1, a2, a3);
+        issue_B(kk0, b0, b1);
+        write_A(a0, a1, a2, a3);
+        write_B(b0, b1);
+      } else {
+        // generic gather: one element at a time (stem conv only)
+        for (int e = tid; e < BMT * BK; e += 256) {
+          const int row = e >> 6, kk = kk0 + (e & 63);
+          __bf16 v = (__bf16)0.f;
+          if (kk < d.K) {
+            int m = m0 + row;
+            if (m >= d.M) m = d.M - 1;
+            int n, oh, ow;
+            decode_m(m, d, n, oh, ow);
+            const int i = kk / SC, rem = kk - i * SC;
+            const int j = rem / d.GC, c = rem - j * d.GC;
+            int ih, iw;
+            if (tap_coord<MODE>(oh, i, d.R, d.stride, d.pad, d.GH, ih) &&
+                tap_coord<MODE>(ow, j, d.S, d.stride, d.pad, d.GW, iw))
+              v = Ag[(((long)n * d.GH + ih) * d.GW + iw) * d.GC + c];
+          }
+          sA[row * LDK + (e & 63)] = v;
+        }
+        for (int e = tid; e < BN * BK; e += 256) {
+          const int row = e >> 6, kk = kk0 + (e & 63);
+          sB[row * LDK + (e & 63)] = (kk < d.K)
+              ? Bg[(long)min(n0 + row, d.N - 1) * d.K + kk] : (__bf16)0.f;
+        }
+      }
+      __syncthreads();
+      mfma_tile();
+      __syncthreads();
+    }
   }
 
   // ---- epilogue: D row = (lane>>4)*4 + reg, col = lane&15 ----------------
