@@ -164,3 +164,84 @@ def test_sampler_shards_are_disjoint():
     i0, i1 = set(iter(s0)), set(iter(s1))
     assert len(i0) == len(i1) == 50
     assert i0.isdisjoint(i1)
+
+
+def _check_torch_ddp_parity(rank, world):
+    """torch's C++ reducer (A/B reference mode) produces the same averaged
+    grads as FlatDDP on the same sharded batch."""
+    from mi355x_ddp.parallel import FlatDDP, wrap_torch_ddp
+    torch.manual_seed(21)
+    net_a = nn.Sequential(nn.Linear(8, 16), nn.ReLU(), nn.Linear(16, 4))
+    torch.manual_seed(21)
+    net_b = nn.Sequential(nn.Linear(8, 16), nn.ReLU(), nn.Linear(16, 4))
+    flat = FlatDDP(net_a)
+    tddp = wrap_torch_ddp(net_b, device_id=None)
+
+    gen = torch.Generator().manual_seed(9)
+    xs = [torch.randn(4, 8, generator=gen) for _ in range(world)]
+    flat.zero_grad_buffer()
+    flat(xs[rank]).pow(2).mean().backward()
+    flat.finalize_backward()
+    tddp(xs[rank]).pow(2).mean().backward()
+    for p, q in zip(net_a.parameters(), net_b.parameters()):
+        assert torch.allclose(p.grad, q.grad, atol=1e-6), \
+            (p.grad - q.grad).abs().max()
+
+
+def _check_grad_accu_matches_large_batch(rank, world):
+    """2 ranks x 2 micro-steps with no_sync == one 4-shard averaged batch."""
+    from mi355x_ddp.parallel import FlatDDP
+    torch.manual_seed(33)
+    model = nn.Linear(6, 3)
+    wrapped = FlatDDP(model)
+    gen = torch.Generator().manual_seed(17)
+    micro = [torch.randn(2, 6, generator=gen) for _ in range(2 * world)]
+
+    wrapped.zero_grad_buffer()
+    with wrapped.no_sync():
+        (wrapped(micro[rank * 2]).pow(2).mean() / 2).backward()
+        wrapped.finalize_backward()
+    (wrapped(micro[rank * 2 + 1]).pow(2).mean() / 2).backward()
+    wrapped.finalize_backward()
+
+    torch.manual_seed(33)
+    ref = nn.Linear(6, 3)
+    sum(ref(m).pow(2).mean() for m in micro).div(2 * world).backward()
+    for p, rp in zip(model.parameters(), ref.parameters()):
+        assert torch.allclose(p.grad, rp.grad, atol=1e-6)
+
+
+def test_torch_ddp_parity(free_port):
+    _run(_check_torch_ddp_parity, free_port)
+
+
+def test_grad_accu_multirank_matches_large_batch(free_port):
+    _run(_check_grad_accu_matches_large_batch, free_port)
+
+
+def test_flat_ddp_bucket_invariants():
+    """Buckets tile the flat buffer exactly: contiguous, disjoint, complete,
+    and every param's grad view lies inside exactly one bucket."""
+    from mi355x_ddp.models import resnet18
+    from mi355x_ddp.parallel import FlatDDP
+    model = FlatDDP(resnet18(), bucket_cap_mb=5)
+    numel = sum(p.numel() for p in model._params)
+    assert model.flat_grads.numel() == numel
+    prev_end = 0
+    for (s, e, ps) in model._buckets:
+        assert s == prev_end and e > s
+        prev_end = e
+    assert prev_end == numel
+    covered = set()
+    for p in model._params:
+        off, view = model._views[p]
+        assert view.data_ptr() == model.flat_grads[off:off + p.numel()].data_ptr()
+        assert p.grad is view and view.shape == p.shape
+        bi = model._param_bucket[p]
+        s, e, ps = model._buckets[bi]
+        assert s <= off and off + p.numel() <= e and p in ps
+        covered.add((off, off + p.numel()))
+    spans = sorted(covered)
+    for (s1, e1), (s2, e2) in zip(spans, spans[1:]):
+        assert e1 <= s2  # disjoint views
+    assert spans[0][0] == 0 and spans[-1][1] == numel

@@ -78,3 +78,21 @@ def test_bench_multirank_contract(tmp_path, free_port):
     assert rec["n_gpus"] == 2 and rec["config"]["parallelism"] == "dp2"
     assert rec["config"]["sync_bn"] is True
     assert rec["value"] > 0 and rec["unit"] == "images/sec"
+
+
+def test_resume_from_checkpoint(tmp_path, free_port):
+    """Train 1 epoch, then resume into a 2-epoch run from the final
+    checkpoint (reference only documented the rank-0 save pattern;
+    resume never existed there)."""
+    _run(tmp_path, "distributed.py", ["--port", str(free_port)])
+    ckpts = sorted((tmp_path / "ckpts").glob("*.pt"))
+    assert ckpts, "no checkpoint written"
+    final = [c for c in ckpts if "final" in c.name][0]
+    r = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "distributed.py")] + COMMON +
+        ["--epochs", "2", "--resume", str(final), "--port", str(free_port + 1)],
+        cwd=tmp_path, env=dict(os.environ, PYTHONPATH=ROOT),
+        capture_output=True, text=True, timeout=420)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "resumed from" in r.stdout
+    assert "Epoch: [1][0/" in r.stdout  # continued at epoch 1, not 0
