@@ -44,6 +44,7 @@ class TrainConfig:
     backend: Optional[str] = None     # None -> nccl(RCCL) when a GPU is visible, else gloo
     sync_bn: bool = True              # reference enables SyncBN in every DDP entry
     bucket_cap_mb: int = 25           # DDP gradient bucket size (tuned for xGMI on HW)
+    dist_timeout_s: int = 600         # collective timeout (a dead rank fails fast)
     use_flat_ddp: bool = True         # MI355X-native flat-bucket reducer (graph-capturable)
 
     # data
@@ -101,6 +102,8 @@ def add_common_args(p: argparse.ArgumentParser) -> argparse.ArgumentParser:
     p.add_argument("--max_train_steps", default=None, type=int,
                    help="cap steps per epoch (smoke runs)")
     p.add_argument("--max_eval_steps", default=None, type=int)
+    p.add_argument("--dist_timeout_s", default=None, type=int,
+                   help="process-group collective timeout in seconds")
     return p
 
 
@@ -128,5 +131,7 @@ def config_from_args(args: argparse.Namespace, **overrides) -> TrainConfig:
         kw["max_train_steps"] = args.max_train_steps
     if getattr(args, "max_eval_steps", None) is not None:
         kw["max_eval_steps"] = args.max_eval_steps
+    if getattr(args, "dist_timeout_s", None) is not None:
+        kw["dist_timeout_s"] = args.dist_timeout_s
     kw.update(overrides)
     return TrainConfig(**kw)

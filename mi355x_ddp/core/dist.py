@@ -8,9 +8,10 @@ env-var init is also supported.
 """
 from __future__ import annotations
 
+import contextlib
 import datetime
 import os
-from typing import Optional
+from typing import Dict, Optional
 
 import torch
 import torch.distributed as dist
@@ -109,3 +110,32 @@ def reduce_mean(tensor: torch.Tensor, nprocs: Optional[int] = None) -> torch.Ten
 def cleanup() -> None:
     if is_dist():
         dist.destroy_process_group()
+
+
+@contextlib.contextmanager
+def count_collectives() -> "Dict[str, int]":
+    """Debug aux (collective launch-count assertions).
+
+    Patches the torch.distributed collectives and yields a dict of call
+    counts, so tests can assert communication BEHAVIOR (e.g. no_sync elides
+    every gradient all-reduce; SyncBN issues exactly one all_reduce per
+    forward) by count rather than only by value.
+    """
+    counts: Dict[str, int] = {}
+    originals = {}
+    for name in ("all_reduce", "all_gather", "broadcast", "barrier",
+                 "reduce_scatter_tensor", "all_gather_into_tensor"):
+        orig = getattr(dist, name)
+        originals[name] = orig
+
+        def make(name, orig):
+            def wrapper(*a, **kw):
+                counts[name] = counts.get(name, 0) + 1
+                return orig(*a, **kw)
+            return wrapper
+        setattr(dist, name, make(name, orig))
+    try:
+        yield counts
+    finally:
+        for name, orig in originals.items():
+            setattr(dist, name, orig)

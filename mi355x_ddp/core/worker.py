@@ -72,7 +72,8 @@ def main_worker(local_rank: int, nprocs: int, cfg: TrainConfig,
     rank = local_rank if rank is None else rank
     if init_pg and nprocs >= 1:
         dist_utils.init_distributed(rank, nprocs, cfg.ip, cfg.port,
-                                    backend=cfg.backend, device_id=local_rank)
+                                    backend=cfg.backend, device_id=local_rank,
+                                    timeout_s=cfg.dist_timeout_s)
     init_seeds(cfg.seed + rank + 1)
     device = torch.device("cuda", local_rank) if torch.cuda.is_available() \
         else torch.device("cpu")

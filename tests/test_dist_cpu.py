@@ -245,3 +245,41 @@ def test_flat_ddp_bucket_invariants():
     for (s1, e1), (s2, e2) in zip(spans, spans[1:]):
         assert e1 <= s2  # disjoint views
     assert spans[0][0] == 0 and spans[-1][1] == numel
+
+
+def _check_collective_counts(rank, world):
+    """no_sync elides EVERY gradient all_reduce; a synced micro-step fires
+    one per bucket; SyncBN fires exactly one all_reduce per fwd and per bwd."""
+    from mi355x_ddp.core.dist import count_collectives
+    from mi355x_ddp.ops.batchnorm import MI355SyncBatchNorm
+    from mi355x_ddp.parallel import FlatDDP
+    torch.manual_seed(1)
+    model = nn.Sequential(nn.Linear(4, 8), nn.Linear(8, 2))
+    wrapped = FlatDDP(model, bucket_cap_mb=1e-5)  # several buckets
+    nbuckets = len(wrapped._buckets)
+    x = torch.randn(2, 4)
+
+    wrapped.zero_grad_buffer()
+    with count_collectives() as counts:
+        with wrapped.no_sync():
+            wrapped(x).sum().backward()
+            wrapped.finalize_backward()
+    assert counts.get("all_reduce", 0) == 0, counts
+
+    with count_collectives() as counts:
+        wrapped(x).sum().backward()
+        wrapped.finalize_backward()
+    assert counts.get("all_reduce", 0) == nbuckets, (counts, nbuckets)
+
+    bn = MI355SyncBatchNorm(4)
+    xb = torch.randn(2, 4, 3, 3, requires_grad=True)
+    with count_collectives() as counts:
+        y = bn(xb)
+    assert counts.get("all_reduce", 0) == 1, counts
+    with count_collectives() as counts:
+        y.sum().backward()
+    assert counts.get("all_reduce", 0) == 1, counts
+
+
+def test_collective_counts(free_port):
+    _run(_check_collective_counts, free_port)
