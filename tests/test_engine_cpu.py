@@ -129,3 +129,62 @@ def test_fp16_scaler_end_to_end():
     scaler.unscale_([model.flat_grads])
     assert not scaler.found_inf
     assert scaler.step(opt)
+
+
+def test_checkpoint_cross_wrap_compat(tmp_path):
+    """A checkpoint saved from a FlatDDP-wrapped model loads into a plain
+    model and vice versa (state dicts are always the unwrapped module's)."""
+    import torch
+    from mi355x_ddp.core.checkpoint import load_checkpoint, save_checkpoint
+    from mi355x_ddp.models import resnet18
+    from mi355x_ddp.parallel import FlatDDP
+
+    torch.manual_seed(0)
+    wrapped = FlatDDP(resnet18())
+    path = save_checkpoint(str(tmp_path), "resnet18", 3, wrapped)
+    assert path is not None
+
+    plain = resnet18()
+    state = load_checkpoint(path, plain)
+    assert state["epoch"] == 3
+    for a, b in zip(plain.state_dict().values(),
+                    wrapped.module.state_dict().values()):
+        assert torch.equal(a, b)
+
+    # and back: plain-saved -> wrapped load
+    path2 = save_checkpoint(str(tmp_path), "resnet18", 4, plain)
+    wrapped2 = FlatDDP(resnet18())
+    load_checkpoint(path2, wrapped2)
+    for a, b in zip(wrapped2.module.state_dict().values(),
+                    plain.state_dict().values()):
+        assert torch.equal(a, b)
+
+
+def test_same_seed_reproduces_losses():
+    """Determinism: identical seeds give identical loss trajectories on CPU
+    (per-rank seeding semantics, reference distributed_mp.py:29-39)."""
+    import torch
+    from mi355x_ddp.config import TrainConfig
+    from mi355x_ddp.core.worker import build_training, init_seeds
+
+    def run():
+        init_seeds(7)
+        cfg = TrainConfig(batch_size=8, amp="fp32", sync_bn=False)
+        device = torch.device("cpu")
+        model, crit, opt, sched, _ = build_training(cfg, device, 1, 0,
+                                                    distributed=True,
+                                                    wrap="flat")
+        gen = torch.Generator().manual_seed(5)
+        losses = []
+        for _ in range(3):
+            x = torch.randn(8, 3, 32, 32, generator=gen)
+            y = torch.randint(0, 100, (8,), generator=gen)
+            model.zero_grad_buffer()
+            loss = crit(model(x), y)
+            loss.backward()
+            model.finalize_backward()
+            opt.step()
+            losses.append(float(loss))
+        return losses
+
+    assert run() == run()
