@@ -130,7 +130,18 @@ def resnet50(num_classes: int = 100) -> ResNet:
     return ResNet(Bottleneck, [3, 4, 6, 3], num_classes)
 
 
-_FACTORIES = {"resnet18": resnet18, "resnet34": resnet34, "resnet50": resnet50}
+def resnet101(num_classes: int = 100) -> ResNet:
+    """Beyond the reference zoo (which stops at resnet50) — same Bottleneck
+    family, deeper stage 3."""
+    return ResNet(Bottleneck, [3, 4, 23, 3], num_classes)
+
+
+def resnet152(num_classes: int = 100) -> ResNet:
+    return ResNet(Bottleneck, [3, 8, 36, 3], num_classes)
+
+
+_FACTORIES = {"resnet18": resnet18, "resnet34": resnet34, "resnet50": resnet50,
+              "resnet101": resnet101, "resnet152": resnet152}
 
 
 def build_model(arch: str, num_classes: int = 100) -> ResNet:

@@ -46,3 +46,13 @@ def test_state_dict_roundtrip():
     x = torch.randn(2, 3, 32, 32)
     m1.eval(), m2.eval()
     assert torch.allclose(m1(x), m2(x))
+
+
+def test_deep_zoo_shapes():
+    from mi355x_ddp.models import resnet101, resnet152
+    for f, blocks in ((resnet101, 33), (resnet152, 50)):
+        m = f(num_classes=100)
+        n_bottleneck = sum(type(x).__name__ == "Bottleneck" for x in m.modules())
+        assert n_bottleneck == blocks
+        y = m(torch.randn(2, 3, 32, 32))
+        assert y.shape == (2, 100)
