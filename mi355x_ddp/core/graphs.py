@@ -47,6 +47,7 @@ class GraphedTrainStep:
                 self._step()
         torch.cuda.current_stream().wait_stream(side)
 
+        self._captured_lr = optimizer.param_groups[0]["lr"]
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph):
             self._step()
@@ -62,7 +63,18 @@ class GraphedTrainStep:
         self.static_loss = loss.detach()
 
     def run(self, images: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
-        """Copy the batch into the static buffers and replay the graph."""
+        """Copy the batch into the static buffers and replay the graph.
+
+        Kernel arguments (including the fused SGD's lr) are frozen at capture
+        time, so an LR-schedule change between replays triggers a re-capture —
+        MultiStepLR changes 3 times in 200 epochs, so this is rare and cheap.
+        """
+        lr = self.optimizer.param_groups[0]["lr"]
+        if lr != self._captured_lr:
+            self._captured_lr = lr
+            self.graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.graph):
+                self._step()
         self.static_img.copy_(images, non_blocking=True)
         self.static_lbl.copy_(labels, non_blocking=True)
         self.graph.replay()
