@@ -43,7 +43,6 @@ def train_one_epoch(model, loader, criterion, optimizer, epoch: int,
     accu = max(1, cfg.grad_accu_steps)
     losses = AverageMeter("Loss", ":.4e")
     step_time = AverageMeter("Time", ":6.3f")
-    last_loss = 0.0
     end = time.time()
 
     for step, (images, labels) in enumerate(loader):
@@ -90,13 +89,14 @@ def train_one_epoch(model, loader, criterion, optimizer, epoch: int,
         else:
             optimizer.step()
 
-        last_loss = float(log_loss)
-        losses.update(last_loss, images.size(0))
         step_time.update(time.time() - end)
         end = time.time()
 
         if step % cfg.log_interval == 0:
+            # the ONLY host sync in the loop: materialise the loss at the
+            # logging interval (a per-step float() would sync every step)
             reduced = dist_utils.reduce_mean(log_loss, nprocs)
+            losses.update(float(reduced), images.size(0))
             if dist_utils.is_main_process():
                 lr = optimizer.param_groups[0]["lr"]
                 print(f"Epoch: [{epoch}][{step}/{len(loader)}]\t"
