@@ -343,7 +343,10 @@ void conv_wgrad_kernel(const __bf16* __restrict__ dy,
     {
       const int cg = grp * 16;
       __bf16 vals[16];
-      if (npq_ok && (FAST || m0 + cg + 16 <= d.M)) {
+      // the row bound also guards FAST: K % 16 == 0 admits K % 64 != 0
+      // (e.g. K=80 via the public conv2d API), where the last tile's 16-wide
+      // float4 loads would run past dy
+      if (npq_ok && m0 + cg + 16 <= d.M) {
         const float4* src = (const float4*)(dy + (long)npq * d.M + m0 + cg);
         *(float4*)&vals[0] = src[0];
         *(float4*)&vals[8] = src[1];

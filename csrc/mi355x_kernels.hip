@@ -713,9 +713,11 @@ __global__ void gap_bwd_kernel(const scalar_t* __restrict__ dy,
 }
 
 // ---------------------------------------------------------------------------
-// Per-row rank of the target class: rank = #{j : logit[j] > logit[target]}.
-// acc@k = mean(rank < k) — equivalent to the reference's topk/eq pipeline
-// (utils/util.py:50-64) for distinct logits, one pass, no sort.
+// Per-row rank of the target class:
+//   rank = #{j : logit[j] > logit[t]} + #{j < t : logit[j] == logit[t]}.
+// acc@k = mean(rank < k). Ties break by smaller class index (a stable
+// descending sort's order — matches the reference's topk/eq pipeline,
+// utils/util.py:50-64, including equal logits). One pass, no sort.
 // ---------------------------------------------------------------------------
 template <typename scalar_t>
 __global__ void class_rank_kernel(const scalar_t* __restrict__ logits,
@@ -724,10 +726,13 @@ __global__ void class_rank_kernel(const scalar_t* __restrict__ logits,
   __shared__ float lds[32];
   const int row = blockIdx.x;
   const scalar_t* xr = logits + (long)row * C;
-  const float tv = (float)xr[target[row]];
+  const int t = (int)target[row];
+  const float tv = (float)xr[t];
   float cnt = 0.f;
-  for (int j = threadIdx.x; j < C; j += blockDim.x)
-    cnt += ((float)xr[j] > tv) ? 1.f : 0.f;
+  for (int j = threadIdx.x; j < C; j += blockDim.x) {
+    const float v = (float)xr[j];
+    cnt += (v > tv || (v == tv && j < t)) ? 1.f : 0.f;
+  }
   cnt = block_reduce_sum(cnt, lds);
   if (threadIdx.x == 0) rank[row] = (int)cnt;
 }
@@ -1260,7 +1265,10 @@ at::Tensor gap_bwd(at::Tensor dy, at::Tensor x_like) {
 
 at::Tensor class_rank(at::Tensor logits, at::Tensor target) {
   CHECK_CUDA(logits); CHECK_CONTIG(logits);
+  CHECK_CUDA(target); CHECK_CONTIG(target);
   TORCH_CHECK(target.scalar_type() == at::kLong, "target must be int64");
+  TORCH_CHECK(target.numel() == logits.size(0),
+              "target must have one entry per logits row");
   const int N = logits.size(0), C = logits.size(1);
   auto rank = at::empty({N}, logits.options().dtype(at::kInt));
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::Half, at::ScalarType::BFloat16,

@@ -43,7 +43,10 @@ class TrainConfig:
     port: int = 23456
     backend: Optional[str] = None     # None -> nccl(RCCL) when a GPU is visible, else gloo
     sync_bn: bool = True              # reference enables SyncBN in every DDP entry
-    bucket_cap_mb: int = 25           # DDP gradient bucket size (tuned for xGMI on HW)
+    # DDP gradient bucket size; None -> parallel.flat_ddp.bucket_cap_for()
+    # picks by world size / model size (xGMI latency-vs-overlap policy)
+    bucket_cap_mb: Optional[float] = None
+    comm_bf16: bool = False           # all-reduce gradients in bf16 (half the wire bytes)
     dist_timeout_s: int = 600         # collective timeout (a dead rank fails fast)
     use_flat_ddp: bool = True         # MI355X-native flat-bucket reducer (graph-capturable)
 

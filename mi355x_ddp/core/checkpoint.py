@@ -35,8 +35,16 @@ def save_checkpoint(ckpt_dir: str, arch: str, epoch: int, model, optimizer=None,
     }
     path = os.path.join(ckpt_dir, f"{arch}_{tag or f'epoch{epoch}'}.pt")
     torch.save(state, path)
+    # {arch}_last.pt is a hardlink to the tagged file — one serialisation
+    # pass, not two (ResNet50+ states are hundreds of MB)
     last = os.path.join(ckpt_dir, f"{arch}_last.pt")
-    torch.save(state, last)
+    try:
+        if os.path.lexists(last):
+            os.remove(last)
+        os.link(path, last)
+    except OSError:  # cross-device / FS without hardlinks
+        import shutil
+        shutil.copyfile(path, last)
     return path
 
 

@@ -23,7 +23,7 @@ from .metrics import AverageMeter, JsonlSink, ProgressMeter, accuracy
 
 
 def _grad_tensors(model):
-    if hasattr(model, "flat_grads"):
+    if getattr(model, "flat_grads", None) is not None:
         return [model.flat_grads]
     return [p.grad for p in model.parameters() if p.grad is not None]
 
@@ -43,6 +43,11 @@ def train_one_epoch(model, loader, criterion, optimizer, epoch: int,
     accu = max(1, cfg.grad_accu_steps)
     losses = AverageMeter("Loss", ":.4e")
     step_time = AverageMeter("Time", ":6.3f")
+    # true epoch-mean loss: accumulated on-device every step (one cheap add
+    # kernel), materialised ONCE at epoch end — no per-step host sync and no
+    # log_interval subsampling bias in the return value
+    epoch_loss_sum = torch.zeros((), device=device)
+    epoch_steps = 0
     end = time.time()
 
     for step, (images, labels) in enumerate(loader):
@@ -89,6 +94,8 @@ def train_one_epoch(model, loader, criterion, optimizer, epoch: int,
         else:
             optimizer.step()
 
+        epoch_loss_sum += log_loss
+        epoch_steps += 1
         step_time.update(time.time() - end)
         end = time.time()
 
@@ -105,7 +112,10 @@ def train_one_epoch(model, loader, criterion, optimizer, epoch: int,
                 if sink is not None:
                     sink.log(kind="train", epoch=epoch, step=step,
                              loss=float(reduced), lr=lr, step_time=step_time.val)
-    return losses.avg
+    if epoch_steps == 0:
+        return 0.0
+    mean = dist_utils.reduce_mean(epoch_loss_sum / epoch_steps, nprocs)
+    return float(mean)
 
 
 @torch.no_grad()

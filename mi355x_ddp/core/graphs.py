@@ -23,7 +23,9 @@ class GraphedTrainStep:
     def __init__(self, model, criterion, optimizer, cfg: TrainConfig,
                  device: torch.device, batch: int, image_size: int = 32,
                  num_classes: int = 100, warmup_iters: int = 3):
-        assert hasattr(model, "flat_grads"), "GraphedTrainStep needs FlatDDP"
+        assert getattr(model, "flat_grads", None) is not None, \
+            "GraphedTrainStep needs FlatDDP with static_grads=True " \
+            "(static grad memory is what makes the capture replayable)"
         self.model = model
         self.criterion = criterion
         self.optimizer = optimizer
@@ -47,10 +49,22 @@ class GraphedTrainStep:
                 self._step()
         torch.cuda.current_stream().wait_stream(side)
 
-        self._captured_lr = optimizer.param_groups[0]["lr"]
+        self._captured_hp = self._hyperparams()
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph):
             self._step()
+
+    def _hyperparams(self):
+        """Every optimizer hyperparameter baked into captured kernel args.
+
+        Snapshot ALL param groups' scalar settings (lr, momentum, weight
+        decay, ...), not just lr — any of them changing between replays means
+        the captured step is stale and must be re-captured.
+        """
+        return tuple(
+            tuple(sorted((k, v) for k, v in g.items()
+                         if isinstance(v, (int, float, bool))))
+            for g in self.optimizer.param_groups)
 
     def _step(self):
         self.model.zero_grad_buffer()
@@ -65,13 +79,19 @@ class GraphedTrainStep:
     def run(self, images: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
         """Copy the batch into the static buffers and replay the graph.
 
-        Kernel arguments (including the fused SGD's lr) are frozen at capture
-        time, so an LR-schedule change between replays triggers a re-capture —
-        MultiStepLR changes 3 times in 200 epochs, so this is rare and cheap.
+        Kernel arguments (the fused SGD's lr/momentum/weight_decay, any
+        added param group's settings) are frozen at capture time, so a change
+        in ANY optimizer hyperparameter triggers a re-capture — MultiStepLR
+        changes 3 times in 200 epochs, so this is rare and cheap.
+
+        world_size>1 contract: the LR schedule (and any other hyperparameter
+        change) must be rank-synchronous — the re-capture replays an RCCL
+        collective, so every rank must re-capture at the same step or the
+        ranks' collectives desynchronise.
         """
-        lr = self.optimizer.param_groups[0]["lr"]
-        if lr != self._captured_lr:
-            self._captured_lr = lr
+        hp = self._hyperparams()
+        if hp != self._captured_hp:
+            self._captured_hp = hp
             self.graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(self.graph):
                 self._step()

@@ -1,8 +1,10 @@
 """Metrics, meters and observability.
 
-Console format matches the reference (utils/util.py:11-48) so output is
-comparable side-by-side; additionally every logged record can go to a JSONL
-sink (the reference's tensorboard_dir knob was dead code, utils/config.py:8).
+Console output is byte-compatible with the reference's meter format
+(utils/util.py:11-48) so runs are comparable side-by-side, but the
+implementation here is its own: meters keep a single (total, n) pair and
+derive everything, and every logged record can also go to a JSONL sink
+(the reference's tensorboard_dir knob was dead code, utils/config.py:8).
 """
 from __future__ import annotations
 
@@ -15,62 +17,79 @@ import torch
 
 
 class AverageMeter:
-    """Running val/sum/count/avg (reference utils/util.py:11-32)."""
+    """Running average with the reference's display format.
+
+    State is one (total, n) accumulator pair plus the last value; ``avg``,
+    ``sum`` and ``count`` are derived properties so the printed string
+    ("{name} {val} ({avg})") matches reference utils/util.py output exactly.
+    """
+
+    __slots__ = ("name", "fmt", "val", "_total", "_n")
 
     def __init__(self, name: str, fmt: str = ":f"):
         self.name = name
         self.fmt = fmt
         self.reset()
 
-    def reset(self):
+    def reset(self) -> None:
         self.val = 0.0
-        self.avg = 0.0
-        self.sum = 0.0
-        self.count = 0
+        self._total = 0.0
+        self._n = 0
 
-    def update(self, val, n: int = 1):
-        self.val = float(val)
-        self.sum += float(val) * n
-        self.count += n
-        self.avg = self.sum / max(1, self.count)
+    def update(self, val, n: int = 1) -> None:
+        v = float(val)
+        self.val = v
+        self._total += v * n
+        self._n += n
 
-    def __str__(self):
-        fmtstr = "{name} {val" + self.fmt + "} ({avg" + self.fmt + "})"
-        return fmtstr.format(**self.__dict__)
+    @property
+    def sum(self) -> float:
+        return self._total
+
+    @property
+    def count(self) -> int:
+        return self._n
+
+    @property
+    def avg(self) -> float:
+        return self._total / self._n if self._n else 0.0
+
+    def __str__(self) -> str:
+        spec = self.fmt.lstrip(":")
+        return f"{self.name} {self.val:{spec}} ({self.avg:{spec}})"
 
 
 class ProgressMeter:
-    """Formatted per-batch progress line (reference utils/util.py:34-48)."""
+    """Per-batch progress line, reference-format ("prefix[ b/B]\\tmeters...")."""
 
-    def __init__(self, num_batches: int, meters: Sequence[AverageMeter], prefix: str = ""):
-        self.batch_fmtstr = self._get_batch_fmtstr(num_batches)
+    def __init__(self, num_batches: int, meters: Sequence[AverageMeter],
+                 prefix: str = ""):
+        width = len(str(num_batches))
+        self._line_head = prefix + "[{:" + str(width) + "d}/" + str(num_batches) + "]"
         self.meters = meters
         self.prefix = prefix
 
-    def display(self, batch: int):
-        entries = [self.prefix + self.batch_fmtstr.format(batch)]
-        entries += [str(meter) for meter in self.meters]
-        print("\t".join(entries), flush=True)
-
-    @staticmethod
-    def _get_batch_fmtstr(num_batches: int) -> str:
-        num_digits = len(str(num_batches // 1))
-        fmt = "{:" + str(num_digits) + "d}"
-        return "[" + fmt + "/" + fmt.format(num_batches) + "]"
+    def display(self, batch: int) -> None:
+        parts = [self._line_head.format(batch)]
+        parts.extend(str(m) for m in self.meters)
+        print("\t".join(parts), flush=True)
 
 
 @torch.no_grad()
 def accuracy(output: torch.Tensor, target: torch.Tensor,
              topk: Iterable[int] = (1, 5)) -> list[torch.Tensor]:
-    """Top-k accuracy in percent (reference utils/util.py:50-64).
+    """Top-k accuracy in percent (capability of reference utils/util.py:50-64).
 
     On GPU this runs the native class_rank kernel: one pass computing
     rank(target) per row — acc@k = mean(rank < k) — instead of ATen's
-    sort-based topk + eq + k reductions.
+    sort-based topk + eq + k reductions. Tie handling: a logit equal to the
+    target's counts as ranked above it only when its class index is smaller,
+    which matches a stable descending sort (torch.topk's observed order).
     """
     from ..ops import _backend
     if _backend.native_enabled(output):
-        rank = _backend.C().class_rank(output.contiguous(), target)
+        rank = _backend.C().class_rank(output.contiguous(),
+                                       target.contiguous())
         return [(rank < k).float().mean().mul(100.0).reshape(1) for k in topk]
     maxk = max(topk)
     batch_size = target.size(0)

@@ -53,7 +53,11 @@ def build_training(cfg: TrainConfig, device: torch.device, world_size: int,
         model = MI355SyncBatchNorm.convert_sync_batchnorm(model)
     if distributed and world_size >= 1 and wrap == "flat":
         model = FlatDDP(model, bucket_cap_mb=cfg.bucket_cap_mb,
-                        overlap=not cfg.hip_graph)
+                        overlap=not cfg.hip_graph,
+                        comm_dtype=torch.bfloat16 if cfg.comm_bf16 else None,
+                        # hipGraph capture needs static grad memory even at
+                        # world 1; otherwise world 1 skips the flat buffer
+                        static_grads=True if cfg.hip_graph else None)
     elif distributed and wrap == "torch":
         model = wrap_torch_ddp(model,
                                device.index if device.type == "cuda" else None,

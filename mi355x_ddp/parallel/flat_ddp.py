@@ -4,43 +4,87 @@ Replaces the reference's dependency on PyTorch's C++ DDP reducer (SURVEY.md
 §2.2 N5: bucketed NCCL all-reduce overlapping backward, no_sync gating).
 Design, chosen for the hardware rather than translated:
 
-- ALL gradients live in ONE contiguous fp32 buffer; every param.grad is a view
-  into it. Autograd accumulates straight into the flat buffer, so there is no
-  bucket-copy pass at all (torch's reducer copies grads into bucket tensors).
+- At world_size > 1, ALL gradients live in ONE contiguous fp32 buffer; every
+  param.grad is a view into it. Autograd accumulates straight into the flat
+  buffer, so there is no bucket-copy pass at all (torch's reducer copies
+  grads into bucket tensors).
+- At world_size == 1 there is nothing to reduce, so no views are installed
+  and grads are set_to_none: autograd's AccumulateGrad then *assigns* the
+  produced gradient instead of launching an add into a pre-existing buffer —
+  that removes ~122 ATen add kernels per ResNet18 step (~0.3 ms of pure
+  launch/elementwise overhead at world 1). `static_grads=True` forces the
+  flat-buffer mode anyway (hipGraph capture needs static grad memory).
 - Buckets are contiguous slices of that buffer in reverse parameter order
-  (backward produces grads roughly last-layer-first), each all-reduced with an
-  async RCCL call from a post-accumulate-grad hook -> communication overlaps
-  the rest of backward over xGMI. Bucket size defaults to cover a ResNet-sized
-  model in a couple of collectives; on 8xMI355X a ring all-reduce is bound by
-  one 153 GB/s xGMI link, so fewer/larger buckets win for models this small
-  (44.9 MB fp32 for ResNet18 — latency-, not bandwidth-bound).
-- `reduce_flat()` all-reduces the whole buffer in one call with no hooks: the
-  hipGraph-capturable path (capture backward + one collective + fused SGD).
-- `no_sync()` suppresses collectives so gradient accumulation micro-steps cost
-  zero communication (reference distributed_gradient_accumulation.py:106).
+  (backward produces grads roughly last-layer-first), each all-reduced with
+  an async RCCL call from a post-accumulate-grad hook -> communication
+  overlaps the rest of backward over xGMI. On 8xMI355X a ring all-reduce is
+  bound by one 153 GB/s xGMI link, so fewer/larger buckets win for models
+  this small (44.9 MB fp32 for ResNet18 — latency-, not bandwidth-bound);
+  `bucket_cap_for()` is the world-size-keyed policy.
+- `comm_dtype=torch.bfloat16` halves the bytes on the wire (apex's fp16
+  gradient path): buckets are cast into a bf16 shadow buffer,
+  all-reduced there, and cast back — accumulation and the optimizer step
+  stay fp32.
+- `reduce_flat()` all-reduces the whole buffer in one call with no hooks:
+  the hipGraph-capturable path (capture backward + one collective + fused
+  SGD).
+- `no_sync()` suppresses collectives so gradient accumulation micro-steps
+  cost zero communication (reference distributed_gradient_accumulation.py:106).
 - Averaging is folded into ONE scale over the flat buffer after the waits
   (cheaper than per-bucket division).
+
+Contract: the overlap path assumes a STATIC graph — every parameter with
+requires_grad receives a gradient on each synchronised backward. A parameter
+that never gets a grad would leave its bucket unfired; `finalize_backward()`
+detects that and raises (listing the parameters) instead of letting the
+ranks hang in mismatched collectives.
 """
 from __future__ import annotations
 
 import contextlib
-from typing import List
+import os
+from typing import List, Optional
 
 import torch
 import torch.distributed as dist
 import torch.nn as nn
 
 
+def bucket_cap_for(world_size: int, total_mb: float) -> float:
+    """Gradient-bucket capacity (MB) for this world size and model size.
+
+    xGMI rationale: each MI355X has 7 point-to-point links at ~153 GB/s; a
+    ring all-reduce serialises per link, so per-bucket wire time is
+    ~2*(W-1)/W * bytes / 153GB/s. For ResNet-sized models (45 MB fp32 /
+    22 MB bf16 of grads) the transfer is latency-dominated: two buckets give
+    backward/comm overlap without paying per-collective launch latency many
+    times. Bigger models get proportionally bigger buckets (fixed ~4
+    collectives per step). `MI355X_BUCKET_CAP_MB` overrides for sweeps.
+    """
+    env = os.environ.get("MI355X_BUCKET_CAP_MB")
+    if env:
+        return float(env)
+    if world_size <= 1:
+        return max(total_mb, 1.0)  # no comm — one nominal bucket
+    if total_mb <= 64.0:
+        return max(total_mb / 2.0, 1.0)   # ResNet18-class: 2 buckets
+    return max(total_mb / 4.0, 25.0)      # larger models: ~4 buckets
+
+
 class FlatDDP(nn.Module):
     def __init__(self, module: nn.Module, process_group=None,
-                 bucket_cap_mb: float = 25.0, overlap: bool = True,
-                 grad_dtype: torch.dtype = torch.float32):
+                 bucket_cap_mb: Optional[float] = None, overlap: bool = True,
+                 grad_dtype: torch.dtype = torch.float32,
+                 comm_dtype: Optional[torch.dtype] = None,
+                 static_grads: Optional[bool] = None):
         super().__init__()
         self.module = module
         self.process_group = process_group
         self.overlap = overlap
         self.require_sync = True
+        self.comm_dtype = comm_dtype
         self._handles: List = []
+        self._fired_buckets = 0
 
         if dist.is_available() and dist.is_initialized():
             self.world_size = dist.get_world_size(process_group)
@@ -51,7 +95,24 @@ class FlatDDP(nn.Module):
         self._params = params
         numel = sum(p.numel() for p in params)
         device = params[0].device if params else torch.device("cpu")
+        total_mb = numel * 4 / (1024 * 1024)
+        if bucket_cap_mb is None:
+            bucket_cap_mb = bucket_cap_for(self.world_size, total_mb)
+
+        self.static_grads = (self.world_size > 1) if static_grads is None \
+            else static_grads
+        if not self.static_grads:
+            # world-1 fast path: no flat buffer, no views — AccumulateGrad
+            # assigns each grad tensor with zero extra kernels
+            self.flat_grads = None
+            self.comm_grads = None
+            self._buckets = []
+            return
+
         self.flat_grads = torch.zeros(numel, dtype=grad_dtype, device=device)
+        self.comm_grads = (
+            torch.zeros(numel, dtype=comm_dtype, device=device)
+            if comm_dtype is not None and comm_dtype != grad_dtype else None)
 
         # reverse registration order ~ backward completion order
         offset = 0
@@ -100,38 +161,76 @@ class FlatDDP(nn.Module):
                     dist.broadcast(t.data, src=0, group=self.process_group)
 
     # -- overlap path ------------------------------------------------------
+    def _launch_bucket(self, bi: int):
+        s, e, _ = self._buckets[bi]
+        if self.comm_grads is not None:
+            self.comm_grads[s:e].copy_(self.flat_grads[s:e])
+            buf = self.comm_grads[s:e]
+        else:
+            buf = self.flat_grads[s:e]
+        h = dist.all_reduce(buf, op=dist.ReduceOp.SUM,
+                            group=self.process_group, async_op=True)
+        self._handles.append(h)
+        self._fired_buckets += 1
+
     def _hook(self, p: torch.Tensor):
         if not self.require_sync or self.world_size == 1:
             return
         bi = self._param_bucket[p]
         self._bucket_pending[bi] -= 1
         if self._bucket_pending[bi] == 0:
-            s, e, ps = self._buckets[bi]
-            h = dist.all_reduce(self.flat_grads[s:e], op=dist.ReduceOp.SUM,
-                                group=self.process_group, async_op=True)
-            self._handles.append(h)
-            self._bucket_pending[bi] = len(ps)
+            self._launch_bucket(bi)
+            self._bucket_pending[bi] = len(self._buckets[bi][2])
 
     def finalize_backward(self):
         """Wait outstanding bucket collectives and average. Call after
         loss.backward() and before optimizer.step()."""
         if self.world_size == 1 or not self.require_sync:
             self._handles.clear()
+            self._fired_buckets = 0
             return
         if not self.overlap:
             self.reduce_flat()
             return
+        if self._fired_buckets != len(self._buckets):
+            # a bucket never filled: some parameter produced no gradient.
+            # Raising here (on every rank, since the graph is the same) beats
+            # the alternative — this rank waiting forever while peers
+            # all-reduce a bucket it never launched.
+            stuck = [name for name, p in self.module.named_parameters()
+                     if p.requires_grad and
+                     self._bucket_pending[self._param_bucket[p]] !=
+                     len(self._buckets[self._param_bucket[p]][2])]
+            raise RuntimeError(
+                "FlatDDP: backward finished with "
+                f"{len(self._buckets) - self._fired_buckets} of "
+                f"{len(self._buckets)} gradient buckets unfired. FlatDDP "
+                "requires a static graph where EVERY requires_grad parameter "
+                "receives a gradient each synchronised step (no conditional "
+                "branches that skip parameters, no find_unused_parameters "
+                "equivalent). Buckets stuck with partial arrivals involve "
+                f"parameters: {stuck or '<none accumulated>'}")
         for h in self._handles:
             h.wait()
         self._handles.clear()
+        self._fired_buckets = 0
+        if self.comm_grads is not None:
+            self.flat_grads.copy_(self.comm_grads)
         self.flat_grads.div_(self.world_size)
 
     # -- graph-capturable path --------------------------------------------
     def reduce_flat(self):
         """One all-reduce over the whole gradient buffer (hipGraph-safe)."""
+        self._fired_buckets = 0
         if self.world_size > 1 and self.require_sync:
-            dist.all_reduce(self.flat_grads, op=dist.ReduceOp.SUM,
-                            group=self.process_group)
+            if self.comm_grads is not None:
+                self.comm_grads.copy_(self.flat_grads)
+                dist.all_reduce(self.comm_grads, op=dist.ReduceOp.SUM,
+                                group=self.process_group)
+                self.flat_grads.copy_(self.comm_grads)
+            else:
+                dist.all_reduce(self.flat_grads, op=dist.ReduceOp.SUM,
+                                group=self.process_group)
             self.flat_grads.div_(self.world_size)
 
     # ----------------------------------------------------------------------
@@ -146,7 +245,11 @@ class FlatDDP(nn.Module):
             self.require_sync = old
 
     def zero_grad_buffer(self):
-        self.flat_grads.zero_()
+        if self.flat_grads is not None:
+            self.flat_grads.zero_()
+        else:
+            for p in self._params:
+                p.grad = None
 
     def zero_grad(self, set_to_none: bool = False):  # keep views alive
         self.zero_grad_buffer()

@@ -15,9 +15,10 @@ import torch.nn as nn
 
 
 def wrap_torch_ddp(model: nn.Module, device_id: Optional[int],
-                   bucket_cap_mb: int = 25, static_graph: bool = False):
+                   bucket_cap_mb: Optional[float] = None,
+                   static_graph: bool = False):
     from torch.nn.parallel import DistributedDataParallel as DDP
-    kwargs = dict(bucket_cap_mb=bucket_cap_mb,
+    kwargs = dict(bucket_cap_mb=25 if bucket_cap_mb is None else bucket_cap_mb,
                   gradient_as_bucket_view=True,
                   static_graph=static_graph)
     if device_id is not None:

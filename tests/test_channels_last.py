@@ -79,7 +79,9 @@ def test_channels_last_grad_views_share_layout():
     net = torch.nn.Sequential(torch.nn.Conv2d(3, 8, 3, padding=1),
                               torch.nn.BatchNorm2d(8))
     net = net.to(memory_format=torch.channels_last)
-    model = FlatDDP(net)
+    # static_grads forces the flat-buffer mode at world 1 (the default there
+    # is the no-view fast path, which has no flat buffer to test)
+    model = FlatDDP(net, static_grads=True)
     conv_w = net[0].weight
     assert conv_w.is_contiguous(memory_format=torch.channels_last)
     assert conv_w.grad.stride() == conv_w.stride()

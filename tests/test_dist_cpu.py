@@ -224,7 +224,7 @@ def test_flat_ddp_bucket_invariants():
     and every param's grad view lies inside exactly one bucket."""
     from mi355x_ddp.models import resnet18
     from mi355x_ddp.parallel import FlatDDP
-    model = FlatDDP(resnet18(), bucket_cap_mb=5)
+    model = FlatDDP(resnet18(), bucket_cap_mb=5, static_grads=True)
     numel = sum(p.numel() for p in model._params)
     assert model.flat_grads.numel() == numel
     prev_end = 0
@@ -283,3 +283,107 @@ def _check_collective_counts(rank, world):
 
 def test_collective_counts(free_port):
     _run(_check_collective_counts, free_port)
+
+
+def _check_bf16_comm_parity(rank, world):
+    """comm_dtype=bf16 all-reduces half the bytes; grads match the fp32-comm
+    path within bf16 rounding."""
+    from mi355x_ddp.parallel import FlatDDP
+    torch.manual_seed(42)
+    model = nn.Sequential(nn.Linear(8, 16), nn.ReLU(), nn.Linear(16, 4))
+    wrapped = FlatDDP(model, bucket_cap_mb=1e-5,
+                      comm_dtype=torch.bfloat16)
+    assert wrapped.comm_grads is not None
+    assert wrapped.comm_grads.dtype == torch.bfloat16
+
+    torch.manual_seed(42)
+    ref_model = nn.Sequential(nn.Linear(8, 16), nn.ReLU(), nn.Linear(16, 4))
+    ref = FlatDDP(ref_model, bucket_cap_mb=1e-5)  # fp32 comm
+
+    gen = torch.Generator().manual_seed(7)
+    xs = [torch.randn(4, 8, generator=gen) for _ in range(world)]
+
+    for w in (wrapped, ref):
+        w.zero_grad_buffer()
+        w(xs[rank]).pow(2).mean().backward()
+        w.finalize_backward()
+
+    for p, rp in zip(model.parameters(), ref_model.parameters()):
+        # bf16 has ~3 decimal digits; grads here are O(0.1)
+        assert torch.allclose(p.grad, rp.grad, atol=2e-3, rtol=2e-2), \
+            (p.grad - rp.grad).abs().max()
+
+    # reduce_flat path too
+    wrapped.zero_grad_buffer()
+    wrapped(xs[rank]).pow(2).mean().backward()
+    wrapped.reduce_flat()
+    gathered = [torch.zeros_like(wrapped.flat_grads) for _ in range(world)]
+    dist.all_gather(gathered, wrapped.flat_grads)
+    assert torch.allclose(gathered[0], gathered[1])  # ranks agree exactly
+
+
+def test_bf16_comm_parity(free_port):
+    _run(_check_bf16_comm_parity, free_port)
+
+
+class _SkipsParam(nn.Module):
+    """Static-graph violation: `unused` never receives a gradient."""
+
+    def __init__(self):
+        super().__init__()
+        self.used = nn.Linear(4, 4)
+        self.unused = nn.Linear(4, 4)
+
+    def forward(self, x):
+        return self.used(x)
+
+
+def _check_unused_param_fails_fast(rank, world):
+    from mi355x_ddp.parallel import FlatDDP
+    torch.manual_seed(0)
+    wrapped = FlatDDP(_SkipsParam(), bucket_cap_mb=1e-5)
+    wrapped.zero_grad_buffer()
+    wrapped(torch.randn(2, 4)).sum().backward()
+    try:
+        wrapped.finalize_backward()
+    except RuntimeError as e:
+        assert "static graph" in str(e) and "unused" in str(e), e
+    else:
+        raise AssertionError("finalize_backward should have raised on the "
+                             "never-filled bucket instead of hanging")
+
+
+def test_unused_param_fails_fast(free_port):
+    """FlatDDP raises (listing the parameters) instead of deadlocking when a
+    parameter never receives a gradient — the static-graph contract."""
+    _run(_check_unused_param_fails_fast, free_port)
+
+
+def test_world1_fast_path_matches_static():
+    """At world 1 FlatDDP installs no grad views (AccumulateGrad assigns,
+    zero add kernels); one optimizer step must match the flat-buffer mode."""
+    from mi355x_ddp.ops import FusedSGD
+    from mi355x_ddp.parallel import FlatDDP
+
+    results = {}
+    for mode in ("fast", "static"):
+        torch.manual_seed(3)
+        net = nn.Sequential(nn.Linear(6, 12), nn.ReLU(), nn.Linear(12, 3))
+        model = FlatDDP(net, static_grads=(mode == "static"))
+        if mode == "fast":
+            assert model.flat_grads is None
+            assert all(p.grad is None for p in net.parameters())
+        else:
+            assert model.flat_grads is not None
+        opt = FusedSGD(model.parameters(), lr=0.1, momentum=0.9,
+                       weight_decay=1e-4)
+        gen = torch.Generator().manual_seed(9)
+        for _ in range(3):
+            x = torch.randn(5, 6, generator=gen)
+            model.zero_grad_buffer()
+            model(x).pow(2).mean().backward()
+            model.finalize_backward()
+            opt.step()
+        results[mode] = [p.detach().clone() for p in net.parameters()]
+    for a, b in zip(results["fast"], results["static"]):
+        assert torch.allclose(a, b, atol=1e-7), (a - b).abs().max()

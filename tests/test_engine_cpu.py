@@ -126,7 +126,9 @@ def test_fp16_scaler_end_to_end():
     loss = crit(model(x), y)
     scaler.scale_loss(loss).backward()
     model.finalize_backward()
-    scaler.unscale_([model.flat_grads])
+    grads = [model.flat_grads] if model.flat_grads is not None else \
+        [p.grad for p in model.parameters() if p.grad is not None]
+    scaler.unscale_(grads)
     assert not scaler.found_inf
     assert scaler.step(opt)
 
