@@ -16,8 +16,13 @@
 // FAST path: when the contiguous channel count is a multiple of BK(=64) a
 // BK-sized reduction chunk lies inside ONE (r,s) filter tap with a contiguous
 // channel run — every A-tile row is one 64-byte-aligned 128 B global read
-// (or zero-fill). This covers every ResNet conv except the 3-channel stem,
-// which takes the generic per-element gather path.
+// (or zero-fill). This covers every ResNet conv except the 3-channel stem.
+//
+// PADC path (the stem): C==3 inputs are padded once per call to 4 channels
+// (x -> NHWC C=4; weight -> [K][K_pad] with K_pad = R*S*4 rounded up to BK,
+// zero-filled), which makes every filter tap a naturally-aligned 8 B dwordx2
+// gather — 4x fewer loads and ~8x less address math than the per-element
+// generic gather that round 1 used for the stem.
 #include <torch/extension.h>
 #include <ATen/ATen.h>
 #include <hip/hip_runtime.h>
@@ -92,7 +97,10 @@ __device__ __forceinline__ void decode_m2(int m, const ConvDims& d,
 // BMT = GEMM-M tile (128 default; 64 doubles the block count for the late
 // small-M stages so they still fill 256 CUs). MI = BMT/32 MFMA row-tiles per
 // wave; EPT = BMT/4 = A-tile elements each thread stages per BK chunk.
-template <int MODE, bool FAST, int BMT = BM>
+// PADC: gathered tensor has GC==4 (channel-padded stem); d.K is R*S*4 rounded
+// up to BK and the B operand is the [N][d.K] zero-padded weight, so every
+// A-tile tap is one aligned dwordx2 and every B row a pair of float4s.
+template <int MODE, bool FAST, int BMT = BM, bool PADC = false>
 __global__ __launch_bounds__(256)
 void conv_igemm_kernel(const __bf16* __restrict__ Ag,
                        const __bf16* __restrict__ Bg,
@@ -195,7 +203,36 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
     }
   };
 
-  if constexpr (FAST && MODE != 2) {
+  if constexpr (PADC) {
+    const int RS = d.R * d.S;
+    for (int kk0 = 0; kk0 < d.K; kk0 += BK) {
+      // A tile: one aligned 8 B dwordx2 per filter tap (4 padded channels)
+      #pragma unroll
+      for (int e = 0; e < EPT / 4; ++e) {
+        const int t4 = a_off + e * 4;
+        const int tap = (kk0 + t4) >> 2;
+        float2 v = {};
+        if (tap < RS) {
+          const int i = tap / d.S, j = tap - (tap / d.S) * d.S;
+          int ih, iw;
+          if (tap_coord<MODE>(a_oh, i, d.R, d.stride, d.pad, d.GH, ih) &&
+              tap_coord<MODE>(a_ow, j, d.S, d.stride, d.pad, d.GW, iw))
+            v = *(const float2*)(Ag +
+                (((long)a_n * d.GH + ih) * d.GW + iw) * 4);
+        }
+        *(float2*)(sA + a_row * LDK + t4) = v;
+      }
+      // B tile: rows are d.K-long (BK multiple) -> always-aligned float4s
+      {
+        float4 b0, b1;
+        issue_B(kk0, b0, b1);
+        write_B(b0, b1);
+      }
+      __syncthreads();
+      mfma_tile();
+      __syncthreads();
+    }
+  } else if constexpr (FAST && MODE != 2) {
     // Register-pipelined: tile k+1's global loads issue before tile k's
     // MFMAs, so HBM/L2 latency overlaps compute; the waits land at the LDS
     // write after the barrier (write-after-barrier form).
@@ -299,19 +336,26 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
 
 // ---------------------------------------------------------------------------
 // wgrad kernel: dw[Kout][RSC] += dy^T[Kout][NPQ-chunk] @ im2col(x)[chunk][RSC]
-// 64x64 output tile per block, 2x2 waves of 32x32; split-K over NPQ chunks
-// with fp32 atomics. Both LDS tiles are written transposed (reduction index
+// 64x64 output tile per block, 2x2 waves of 32x32; split-K over NPQ chunks.
+// Each grid.z slice accumulates ITS chunks in registers and stores a private
+// fp32 partial slab (no atomics — round 1's fp32 atomicAdd contention was the
+// main wgrad cost at small tm*tn); wgrad_reduce_kernel sums the slabs and
+// emits the bf16 channels_last weight grad in one pass. DIRECT (splits==1,
+// no channel padding): the slab and reduce collapse away — the kernel writes
+// bf16 dw itself. Both LDS tiles are written transposed (reduction index
 // contiguous per row) so MFMA fragment reads are 16-byte ds_read_b128.
+// PADC: x has 4 padded channels (stem) — per-tap dwordx2 gathers.
 // ---------------------------------------------------------------------------
-template <bool FAST>
+template <bool FAST, bool PADC = false, bool DIRECT = false>
 __global__ __launch_bounds__(256)
 void conv_wgrad_kernel(const __bf16* __restrict__ dy,
                        const __bf16* __restrict__ x,
-                       float* __restrict__ dw, ConvDims d) {
+                       float* __restrict__ dwp,
+                       __bf16* __restrict__ dwb, ConvDims d) {
   // d: M = Kout, N = R*S*C, K = Nb*P*Q; OH/OW = P,Q; GH/GW/GC = H,W,C
   // 64(Kout) x 128(rsc) tile: each loaded byte feeds twice the MFMA work of
-  // the 64x64 tile, and the split-K atomics halve. FAST needs C % 64 == 0 so
-  // every 32-wide rsc sub-chunk stays inside one (r,s) filter tap.
+  // the 64x64 tile. FAST needs C % 64 == 0 so every 32-wide rsc sub-chunk
+  // stays inside one (r,s) filter tap.
   __shared__ __bf16 sA[64 * LDK];   // [kout][npq]
   __shared__ __bf16 sB[128 * LDK];  // [rsc][npq]
 
@@ -364,7 +408,24 @@ void conv_wgrad_kernel(const __bf16* __restrict__ dy,
     {
       const int cg = grp * 32;     // this thread's 32-col sub-chunk
       __bf16 vals[32];
-      if (FAST) {
+      if (PADC) {
+        // stem: 8 taps of 4 padded channels each, aligned dwordx2 gathers
+        #pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int nn = n0 + cg + e * 4;
+          float2 v = {};
+          if (npq_ok && nn < d.N) {
+            const int tap = nn >> 2;
+            const int i = tap / d.S, j = tap - (tap / d.S) * d.S;
+            const int ih = xp * d.stride + i - d.pad;
+            const int iw = xq * d.stride + j - d.pad;
+            if (ih >= 0 && ih < d.GH && iw >= 0 && iw < d.GW)
+              v = *(const float2*)(x +
+                  (((long)xn * d.GH + ih) * d.GW + iw) * 4);
+          }
+          *(float2*)&vals[e * 4] = v;
+        }
+      } else if (FAST) {
         const int nn = n0 + cg;
         const int i = nn / SC;
         const int rem = nn - i * SC;
@@ -426,6 +487,7 @@ void conv_wgrad_kernel(const __bf16* __restrict__ dy,
 
   const int dm = (lane >> 4) * 4;
   const int dn = lane & 15;
+  float* slab = DIRECT ? nullptr : dwp + (long)blockIdx.z * d.M * d.N;
   #pragma unroll
   for (int mi = 0; mi < 2; ++mi)
     #pragma unroll
@@ -435,9 +497,35 @@ void conv_wgrad_kernel(const __bf16* __restrict__ dy,
       #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int m = m0 + wm + mi * 16 + dm + r;
-        if (m < d.M) atomicAdd(&dw[(long)m * d.N + n], acc[mi][ni][r]);
+        if (m >= d.M) continue;
+        if (DIRECT) dwb[(long)m * d.N + n] = (__bf16)acc[mi][ni][r];
+        else slab[(long)m * d.N + n] = acc[mi][ni][r];
       }
     }
+}
+
+// Sum the split-K partial slabs and emit the bf16 channels_last weight grad:
+// out[k][tap*Cout + c] (= memory layout of channels_last (K, Cout, R, S)).
+// Cpad==Cout -> identity column map; the stem maps Cpad=4 -> Cout=3 and drops
+// the zero-pad channel and any tap >= R*S.
+__global__ void wgrad_reduce_kernel(const float* __restrict__ part,
+                                    __bf16* __restrict__ dw,
+                                    long MN, int N, int Nout, int splits,
+                                    int Cpad, int Cout) {
+  for (long o = (long)blockIdx.x * blockDim.x + threadIdx.x; o < MN;
+       o += (long)gridDim.x * blockDim.x) {
+    const long m = o / N;
+    const int n = (int)(o - m * N);
+    int nout = n;
+    if (Cpad != Cout) {
+      const int tap = n / Cpad, c = n - tap * Cpad;
+      if (c >= Cout || tap * Cout >= Nout) continue;
+      nout = tap * Cout + c;
+    }
+    float s = 0.f;
+    for (int z = 0; z < splits; ++z) s += part[(long)z * MN + o];
+    dw[m * Nout + nout] = (__bf16)s;
+  }
 }
 
 // wT[c][i][j][k] = w[k][R-1-i][S-1-j][c] — the 180°-rotated transposed filter
@@ -455,6 +543,48 @@ __global__ void build_wT_kernel(const __bf16* __restrict__ w,
     const int i = (int)(t % R);
     const int c = (int)(t / R);
     wT[o] = w[(((long)k * R + (R - 1 - i)) * S + (S - 1 - j)) * C + c];
+  }
+}
+
+// x4[p][0..3] = {x[p][0..2], 0} — NHWC C=3 -> C=4 (stem fast path). Two
+// pixels per thread: 3 aligned dword loads -> 2 aligned dwordx2 stores.
+__global__ void pad_c3_to_c4_kernel(const __bf16* __restrict__ x,
+                                    __bf16* __restrict__ x4, long npix) {
+  typedef __attribute__((ext_vector_type(2))) float f32x2;
+  const long pairs = (npix + 1) / 2;
+  for (long o = (long)blockIdx.x * blockDim.x + threadIdx.x; o < pairs;
+       o += (long)gridDim.x * blockDim.x) {
+    const long p = o * 2;
+    __bf16 v[8] = {};
+    if (p + 2 <= npix) {
+      const float* src = (const float*)(x + p * 3);   // 12 B, 4 B aligned
+      *(float*)&v[0] = src[0];
+      *(float*)&v[2] = src[1];   // v[2],v[3] = {x[p][2], x[p+1][0]}
+      *(float*)&v[6] = src[2];
+      v[4] = v[3]; v[5] = v[6]; v[6] = v[7]; v[3] = (__bf16)0.f;
+      v[7] = (__bf16)0.f;
+    } else {  // odd tail pixel
+      for (int c = 0; c < 3; ++c) v[c] = x[p * 3 + c];
+    }
+    float2* dst = (float2*)(x4 + p * 4);
+    dst[0] = *(float2*)&v[0];
+    if (p + 2 <= npix) dst[1] = *(float2*)&v[4];
+  }
+}
+
+// w4[k][tap*4 + c] = w[k][tap*3 + c] for c<3, tap<R*S; else 0.
+// Kpad = R*S*4 rounded up to BK. w is channels_last (K,3,R,S) = [K][R][S][3].
+__global__ void pad_w_c4_kernel(const __bf16* __restrict__ w,
+                                __bf16* __restrict__ w4,
+                                int K, int RS, int Kpad) {
+  const long total = (long)K * Kpad;
+  for (long o = (long)blockIdx.x * blockDim.x + threadIdx.x; o < total;
+       o += (long)gridDim.x * blockDim.x) {
+    const int k = (int)(o / Kpad);
+    const int n = (int)(o - (long)k * Kpad);
+    const int tap = n >> 2, c = n & 3;
+    w4[o] = (tap < RS && c < 3) ? w[((long)k * RS + tap) * 3 + c]
+                                : (__bf16)0.f;
   }
 }
 
@@ -488,9 +618,20 @@ at::Tensor conv_build_wT(at::Tensor w) {
   return wT;
 }
 
+// tile: 0 = size heuristic, 64 or 128 = force that GEMM-M tile (the
+// autotune cache in ops/conv.py measures both and pins the winner per shape).
+namespace {
+inline int pick_bmt(const ConvDims& d, long tile) {
+  if (tile == 64 || tile == 128) return (int)tile;
+  const long tiles128 = (long)((d.M + BM - 1) / BM) * ((d.N + BN - 1) / BN);
+  return tiles128 < 384 ? 64 : 128;  // small-M: halve tile, double blocks
+}
+}  // namespace
+
 // out (N,K,P,Q) channels_last <- x (N,C,H,W) channels_last, w (K,C,R,S)
 // channels_last (memory [K][R][S][C]).
-at::Tensor conv_fwd_igemm(at::Tensor x, at::Tensor w, long stride, long pad) {
+at::Tensor conv_fwd_igemm(at::Tensor x, at::Tensor w, long stride, long pad,
+                          long tile) {
   check_nhwc_bf16(x, "x"); check_nhwc_bf16(w, "w");
   const int Nb = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   const int K = w.size(0), R = w.size(2), S = w.size(3);
@@ -499,31 +640,57 @@ at::Tensor conv_fwd_igemm(at::Tensor x, at::Tensor w, long stride, long pad) {
   const int Q = (W + 2 * (int)pad - S) / (int)stride + 1;
   auto out = at::empty({Nb, K, P, Q},
                        x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  auto stream = conv_stream();
+
+  if (C == 3) {
+    // stem fast path: pad x and w to 4 channels, tap-wise dwordx2 gathers
+    const int RS = R * S;
+    const int Kpad = ((RS * 4 + BK - 1) / BK) * BK;
+    const long npix = (long)Nb * H * W;
+    auto x4 = at::empty({npix * 4}, x.options());
+    auto w4 = at::empty({(long)K * Kpad}, w.options());
+    {
+      const int blocks = (int)std::min((npix / 2 + 255) / 256, (long)4096);
+      hipLaunchKernelGGL(pad_c3_to_c4_kernel, dim3(std::max(blocks, 1)),
+                         dim3(256), 0, stream, bf16_ptr(x),
+                         reinterpret_cast<__bf16*>(x4.data_ptr()), npix);
+      const long wtotal = (long)K * Kpad;
+      hipLaunchKernelGGL(pad_w_c4_kernel,
+                         dim3((int)std::min((wtotal + 255) / 256, (long)1024)),
+                         dim3(256), 0, stream, bf16_ptr(w),
+                         reinterpret_cast<__bf16*>(w4.data_ptr()), K, RS, Kpad);
+    }
+    ConvDims d{Nb, P, Q, H, W, 4, R, S, (int)stride, (int)pad,
+               Nb * P * Q, K, Kpad};
+    const int bmt = pick_bmt(d, tile);
+    const dim3 grid((d.N + BN - 1) / BN, (d.M + bmt - 1) / bmt);
+    auto* kern = bmt == 64 ? conv_igemm_kernel<0, false, 64, true>
+                           : conv_igemm_kernel<0, false, 128, true>;
+    hipLaunchKernelGGL(kern, grid, dim3(256), 0, stream,
+                       reinterpret_cast<const __bf16*>(x4.data_ptr()),
+                       reinterpret_cast<const __bf16*>(w4.data_ptr()),
+                       reinterpret_cast<__bf16*>(out.data_ptr()), d);
+    return out;
+  }
+
   ConvDims d{Nb, P, Q, H, W, C, R, S, (int)stride, (int)pad,
              Nb * P * Q, K, R * S * C};
   const bool fast = (C % BK == 0);
-  const long tiles128 = (long)((d.M + BM - 1) / BM) * ((d.N + BN - 1) / BN);
-  if (tiles128 < 384) {  // small-M late stages: halve the tile, double blocks
-    const dim3 grid((d.N + BN - 1) / BN, (d.M + 63) / 64);
-    auto* kern = fast ? conv_igemm_kernel<0, true, 64>
-                      : conv_igemm_kernel<0, false, 64>;
-    hipLaunchKernelGGL(kern, grid, dim3(256), 0, conv_stream(),
-                       bf16_ptr(x), bf16_ptr(w),
-                       reinterpret_cast<__bf16*>(out.data_ptr()), d);
-  } else {
-    const dim3 grid((d.N + BN - 1) / BN, (d.M + BM - 1) / BM);
-    auto* kern = fast ? conv_igemm_kernel<0, true> : conv_igemm_kernel<0, false>;
-    hipLaunchKernelGGL(kern, grid, dim3(256), 0, conv_stream(),
-                       bf16_ptr(x), bf16_ptr(w),
-                       reinterpret_cast<__bf16*>(out.data_ptr()), d);
-  }
+  const int bmt = pick_bmt(d, tile);
+  const dim3 grid((d.N + BN - 1) / BN, (d.M + bmt - 1) / bmt);
+  auto* kern = bmt == 64
+      ? (fast ? conv_igemm_kernel<0, true, 64> : conv_igemm_kernel<0, false, 64>)
+      : (fast ? conv_igemm_kernel<0, true> : conv_igemm_kernel<0, false>);
+  hipLaunchKernelGGL(kern, grid, dim3(256), 0, stream,
+                     bf16_ptr(x), bf16_ptr(w),
+                     reinterpret_cast<__bf16*>(out.data_ptr()), d);
   return out;
 }
 
 // dx (N,C,H,W) channels_last <- dy (N,K,P,Q) channels_last,
 // wT (C,R,S,K) CONTIGUOUS with taps 180°-rotated (built by the Python side).
 at::Tensor conv_dgrad_igemm(at::Tensor dy, at::Tensor wT, long H, long W,
-                            long stride, long pad) {
+                            long stride, long pad, long tile) {
   check_nhwc_bf16(dy, "dy");
   TORCH_CHECK(wT.is_cuda() && wT.scalar_type() == at::ScalarType::BFloat16 &&
               wT.is_contiguous(), "wT must be contiguous bf16");
@@ -540,55 +707,98 @@ at::Tensor conv_dgrad_igemm(at::Tensor dy, at::Tensor wT, long H, long W,
     // class-local M; wrong-parity taps are skipped inside the kernel.
     ConvDims d2 = d;
     d2.M = Nb * (int)(H / 2) * (int)(W / 2);
-    const long tiles = (long)((d2.M + BM - 1) / BM) * ((d2.N + BN - 1) / BN) * 4;
-    if (tiles < 384) {
-      const dim3 grid((d2.N + BN - 1) / BN, (d2.M + 63) / 64, 4);
-      hipLaunchKernelGGL((conv_igemm_kernel<2, true, 64>), grid, dim3(256), 0,
-                         conv_stream(), bf16_ptr(dy), bf16_ptr(wT),
-                         reinterpret_cast<__bf16*>(dx.data_ptr()), d2);
-    } else {
-      const dim3 grid((d2.N + BN - 1) / BN, (d2.M + BM - 1) / BM, 4);
-      hipLaunchKernelGGL((conv_igemm_kernel<2, true, 128>), grid, dim3(256), 0,
-                         conv_stream(), bf16_ptr(dy), bf16_ptr(wT),
-                         reinterpret_cast<__bf16*>(dx.data_ptr()), d2);
+    // 1x1 s2: only class (0,0) receives anything — 3/4 of dx is zero.
+    // memset + one class beats 4 classes computing mostly nothing.
+    const int nclass = (R == 1 && S == 1 && pad == 0) ? 1 : 4;
+    if (nclass == 1) {
+      const hipError_t err = hipMemsetAsync(
+          dx.data_ptr(), 0, dx.numel() * dx.element_size(), conv_stream());
+      TORCH_CHECK(err == hipSuccess, "hipMemsetAsync failed: ",
+                  hipGetErrorString(err));
     }
+    const long t128 = (long)((d2.M + BM - 1) / BM) *
+                      ((d2.N + BN - 1) / BN) * nclass;
+    const int bmt = (tile == 64 || tile == 128) ? (int)tile
+                                                : (t128 < 384 ? 64 : 128);
+    const dim3 grid((d2.N + BN - 1) / BN, (d2.M + bmt - 1) / bmt, nclass);
+    auto* kern = bmt == 64 ? conv_igemm_kernel<2, true, 64>
+                           : conv_igemm_kernel<2, true, 128>;
+    hipLaunchKernelGGL(kern, grid, dim3(256), 0,
+                       conv_stream(), bf16_ptr(dy), bf16_ptr(wT),
+                       reinterpret_cast<__bf16*>(dx.data_ptr()), d2);
     return dx;
   }
-  const long tiles128 = (long)((d.M + BM - 1) / BM) * ((d.N + BN - 1) / BN);
-  if (tiles128 < 384) {
-    const dim3 grid((d.N + BN - 1) / BN, (d.M + 63) / 64);
-    auto* kern = fast ? conv_igemm_kernel<1, true, 64>
-                      : conv_igemm_kernel<1, false, 64>;
-    hipLaunchKernelGGL(kern, grid, dim3(256), 0, conv_stream(),
-                       bf16_ptr(dy), bf16_ptr(wT),
-                       reinterpret_cast<__bf16*>(dx.data_ptr()), d);
-  } else {
-    const dim3 grid((d.N + BN - 1) / BN, (d.M + BM - 1) / BM);
-    auto* kern = fast ? conv_igemm_kernel<1, true> : conv_igemm_kernel<1, false>;
-    hipLaunchKernelGGL(kern, grid, dim3(256), 0, conv_stream(),
-                       bf16_ptr(dy), bf16_ptr(wT),
-                       reinterpret_cast<__bf16*>(dx.data_ptr()), d);
-  }
+  const int bmt = pick_bmt(d, tile);
+  const dim3 grid((d.N + BN - 1) / BN, (d.M + bmt - 1) / bmt);
+  auto* kern = bmt == 64
+      ? (fast ? conv_igemm_kernel<1, true, 64> : conv_igemm_kernel<1, false, 64>)
+      : (fast ? conv_igemm_kernel<1, true> : conv_igemm_kernel<1, false>);
+  hipLaunchKernelGGL(kern, grid, dim3(256), 0, conv_stream(),
+                     bf16_ptr(dy), bf16_ptr(wT),
+                     reinterpret_cast<__bf16*>(dx.data_ptr()), d);
   return dx;
 }
 
-// dw fp32 [K][R*S*C] (= channels_last layout of (K,C,R,S)) <- dy, x.
+// dw bf16 (K,C,R,S) channels_last <- dy, x. Two-stage split-K: private fp32
+// partial slabs per grid.z slice + one reduce pass (no atomics); splits==1
+// non-stem collapses to a DIRECT bf16 store. `splits`: 0 = heuristic.
 at::Tensor conv_wgrad_igemm(at::Tensor dy, at::Tensor x, long R, long S,
-                            long stride, long pad) {
+                            long stride, long pad, long splits_arg) {
   check_nhwc_bf16(dy, "dy"); check_nhwc_bf16(x, "x");
   const int Nb = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   const int K = dy.size(1), P = dy.size(2), Q = dy.size(3);
-  auto dw = at::zeros({K, (long)(R * S * C)}, x.options().dtype(at::kFloat));
-  ConvDims d{Nb, P, Q, H, W, C, (int)R, (int)S, (int)stride, (int)pad,
-             K, (int)(R * S * C), Nb * P * Q};
+  auto stream = conv_stream();
+  auto dw = at::empty({K, C, R, S},
+                      x.options().memory_format(at::MemoryFormat::ChannelsLast));
+
+  const bool stem = (C == 3);
+  int Cpad = C, Npad = (int)(R * S * C);
+  at::Tensor x4;
+  const __bf16* xp = bf16_ptr(x);
+  if (stem) {
+    const long npix = (long)Nb * H * W;
+    x4 = at::empty({npix * 4}, x.options());
+    const int blocks = (int)std::min((npix / 2 + 255) / 256, (long)4096);
+    hipLaunchKernelGGL(pad_c3_to_c4_kernel, dim3(std::max(blocks, 1)),
+                       dim3(256), 0, stream, bf16_ptr(x),
+                       reinterpret_cast<__bf16*>(x4.data_ptr()), npix);
+    xp = reinterpret_cast<const __bf16*>(x4.data_ptr());
+    Cpad = 4;
+    Npad = (int)(R * S * 4);
+  }
+
+  ConvDims d{Nb, P, Q, H, W, Cpad, (int)R, (int)S, (int)stride, (int)pad,
+             K, Npad, Nb * P * Q};
   const int tm = (d.M + 63) / 64, tn = (d.N + 127) / 128;
   const int nchunks = (d.K + BK - 1) / BK;
-  int splits = 640 / (tm * tn);
-  splits = std::max(1, std::min(splits, nchunks));
+  int splits = (splits_arg > 0) ? (int)splits_arg : 768 / (tm * tn);
+  // bound the partial-slab workspace to ~96 MB
+  const long max_ws = 96L * 1024 * 1024 / ((long)d.M * d.N * 4);
+  splits = std::max(1, (int)std::min({(long)splits, (long)nchunks,
+                                      std::max(max_ws, 1L), 256L}));
   const dim3 grid(tn, tm, splits);
   const bool fast = (C % BK == 0) && (K % 16 == 0);
-  auto* kern = fast ? conv_wgrad_kernel<true> : conv_wgrad_kernel<false>;
-  hipLaunchKernelGGL(kern, grid, dim3(256), 0, conv_stream(),
-                     bf16_ptr(dy), bf16_ptr(x), dw.data_ptr<float>(), d);
+
+  if (splits == 1 && !stem) {
+    auto* kern = fast ? conv_wgrad_kernel<true, false, true>
+                      : conv_wgrad_kernel<false, false, true>;
+    hipLaunchKernelGGL(kern, grid, dim3(256), 0, stream, bf16_ptr(dy), xp,
+                       nullptr, reinterpret_cast<__bf16*>(dw.data_ptr()), d);
+    return dw;
+  }
+
+  auto part = at::empty({(long)splits * d.M * d.N},
+                        x.options().dtype(at::kFloat));
+  auto* kern = stem ? conv_wgrad_kernel<false, true, false>
+                    : (fast ? conv_wgrad_kernel<true, false, false>
+                            : conv_wgrad_kernel<false, false, false>);
+  hipLaunchKernelGGL(kern, grid, dim3(256), 0, stream, bf16_ptr(dy), xp,
+                     part.data_ptr<float>(), nullptr, d);
+  const long MN = (long)d.M * d.N;
+  hipLaunchKernelGGL(wgrad_reduce_kernel,
+                     dim3((int)std::min((MN + 255) / 256, (long)4096)),
+                     dim3(256), 0, stream, part.data_ptr<float>(),
+                     reinterpret_cast<__bf16*>(dw.data_ptr()), MN, d.N,
+                     (int)(R * S * C), splits, Cpad, C);
   return dw;
 }

@@ -1283,21 +1283,29 @@ at::Tensor class_rank(at::Tensor logits, at::Tensor target) {
 
 // csrc/conv_igemm.hip — implicit-GEMM MFMA convolution (NHWC bf16)
 at::Tensor conv_build_wT(at::Tensor w);
-at::Tensor conv_fwd_igemm(at::Tensor x, at::Tensor w, long stride, long pad);
+at::Tensor conv_fwd_igemm(at::Tensor x, at::Tensor w, long stride, long pad,
+                          long tile);
 at::Tensor conv_dgrad_igemm(at::Tensor dy, at::Tensor wT, long H, long W,
-                            long stride, long pad);
+                            long stride, long pad, long tile);
 at::Tensor conv_wgrad_igemm(at::Tensor dy, at::Tensor x, long R, long S,
-                            long stride, long pad);
+                            long stride, long pad, long splits);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_build_wT", &conv_build_wT,
         "build rotated/transposed filter for dgrad (one launch)");
   m.def("conv_fwd_igemm", &conv_fwd_igemm,
-        "implicit-GEMM conv forward (NHWC bf16, MFMA)");
+        "implicit-GEMM conv forward (NHWC bf16, MFMA); tile 0=auto|64|128",
+        py::arg("x"), py::arg("w"), py::arg("stride"), py::arg("pad"),
+        py::arg("tile") = 0);
   m.def("conv_dgrad_igemm", &conv_dgrad_igemm,
-        "implicit-GEMM conv input-grad (NHWC bf16, MFMA)");
+        "implicit-GEMM conv input-grad (NHWC bf16, MFMA); tile 0=auto|64|128",
+        py::arg("dy"), py::arg("wT"), py::arg("H"), py::arg("W"),
+        py::arg("stride"), py::arg("pad"), py::arg("tile") = 0);
   m.def("conv_wgrad_igemm", &conv_wgrad_igemm,
-        "implicit-GEMM conv weight-grad (NHWC bf16, MFMA, split-K fp32)");
+        "implicit-GEMM conv weight-grad -> bf16 channels_last (K,C,R,S); "
+        "two-stage split-K, no atomics; splits 0=auto",
+        py::arg("dy"), py::arg("x"), py::arg("R"), py::arg("S"),
+        py::arg("stride"), py::arg("pad"), py::arg("splits") = 0);
   m.def("bn_stats", &bn_stats, "per-channel sum/sqsum (NCHW)");
   m.def("bn_stats_packed", &bn_stats_packed,
         "per-channel {sum,sqsum} packed [2C], no-atomic NHWC v2");

@@ -7,10 +7,17 @@ run when the tensor is a bf16 channels_last CUDA tensor; anything else
 (CPU tests, fp32 mode, exotic shapes) runs torch's conv so the same module
 is testable everywhere. ``MI355X_NATIVE_CONV=0`` forces the torch path for
 parity A/B.
+
+Autotune (the ``cudnn.benchmark=True`` equivalent the reference relies on,
+distributed.py:48): per conv shape, the GEMM-M tile (64 vs 128) for
+fwd/dgrad and the split-K factor for wgrad are measured once at first use
+with hip events and cached for the rest of the process. ``MI355X_AUTOTUNE=0``
+falls back to the static size heuristic inside the kernels.
 """
 from __future__ import annotations
 
 import os
+from typing import Callable, Dict, Sequence, Tuple
 
 import torch
 import torch.nn as nn
@@ -21,6 +28,10 @@ from . import _backend
 
 def native_conv_wanted() -> bool:
     return os.environ.get("MI355X_NATIVE_CONV", "1") == "1"
+
+
+def autotune_wanted() -> bool:
+    return os.environ.get("MI355X_AUTOTUNE", "1") == "1"
 
 
 def _native_ok(x: torch.Tensor, weight: torch.Tensor, stride, padding,
@@ -41,6 +52,46 @@ def _native_ok(x: torch.Tensor, weight: torch.Tensor, stride, padding,
     return weight.shape[2] == weight.shape[3]
 
 
+# --------------------------------------------------------------------------
+# Per-shape launch-config cache (kernel-selection cache a la cudnn.benchmark)
+# --------------------------------------------------------------------------
+
+_TUNE_CACHE: Dict[Tuple, int] = {}
+
+
+def _measure_ms(fn: Callable[[], None], iters: int = 4) -> float:
+    fn()  # allocator warm-up
+    torch.cuda.synchronize()
+    start = torch.cuda.Event(enable_timing=True)
+    stop = torch.cuda.Event(enable_timing=True)
+    start.record()
+    for _ in range(iters):
+        fn()
+    stop.record()
+    stop.synchronize()
+    return start.elapsed_time(stop)
+
+
+def _tuned_choice(key: Tuple, candidates: Sequence[int],
+                  run: Callable[[int], None]) -> int:
+    """Cached fastest candidate for this shape key (0 = kernel heuristic)."""
+    if key in _TUNE_CACHE:
+        return _TUNE_CACHE[key]
+    if not autotune_wanted() or torch.cuda.is_current_stream_capturing():
+        return 0
+    best, best_ms = 0, float("inf")
+    for c in candidates:
+        ms = _measure_ms(lambda: run(c))
+        if ms < best_ms:
+            best, best_ms = c, ms
+    _TUNE_CACHE[key] = best
+    return best
+
+
+def clear_autotune_cache() -> None:
+    _TUNE_CACHE.clear()
+
+
 class _ConvIGEMM(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, stride: int, padding: int):
@@ -48,26 +99,40 @@ class _ConvIGEMM(torch.autograd.Function):
         weight = weight.contiguous(memory_format=torch.channels_last)
         ctx.save_for_backward(x, weight)
         ctx.stride, ctx.padding = stride, padding
-        return _backend.C().conv_fwd_igemm(x, weight, stride, padding)
+        C = _backend.C()
+        key = ("fwd", x.shape, weight.shape, stride, padding)
+        tile = _tuned_choice(
+            key, (64, 128),
+            lambda t: C.conv_fwd_igemm(x, weight, stride, padding, t))
+        return C.conv_fwd_igemm(x, weight, stride, padding, tile)
 
     @staticmethod
     def backward(ctx, dy):
         x, weight = ctx.saved_tensors
         dy = dy.contiguous(memory_format=torch.channels_last)
         dx = dw = None
+        C = _backend.C()
         if ctx.needs_input_grad[0]:
             # 180°-rotated, (C,R,S,K)-transposed filter for the dgrad GEMM
-            wT = _backend.C().conv_build_wT(weight)
-            dx = _backend.C().conv_dgrad_igemm(dy, wT, x.shape[2], x.shape[3],
-                                               ctx.stride, ctx.padding)
+            wT = C.conv_build_wT(weight)
+            key = ("dgrad", dy.shape, wT.shape, ctx.stride, ctx.padding)
+            tile = _tuned_choice(
+                key, (64, 128),
+                lambda t: C.conv_dgrad_igemm(dy, wT, x.shape[2], x.shape[3],
+                                             ctx.stride, ctx.padding, t))
+            dx = C.conv_dgrad_igemm(dy, wT, x.shape[2], x.shape[3],
+                                    ctx.stride, ctx.padding, tile)
         if ctx.needs_input_grad[1]:
             R = weight.shape[2]
-            dw_f32 = _backend.C().conv_wgrad_igemm(dy, x, R, R,
-                                                   ctx.stride, ctx.padding)
-            # fp32 [K][R*S*C] is exactly the channels_last layout of (K,C,R,S)
-            dw_view = dw_f32.view(weight.shape[0], R, R, weight.shape[1]) \
-                            .permute(0, 3, 1, 2)
-            dw = torch.empty_like(weight).copy_(dw_view)
+            key = ("wgrad", dy.shape, x.shape, R, ctx.stride, ctx.padding)
+            splits = _tuned_choice(
+                key, (0, 1, 32, 128),
+                lambda s: C.conv_wgrad_igemm(dy, x, R, R, ctx.stride,
+                                             ctx.padding, s))
+            # kernel emits bf16 (K,C,R,S) channels_last directly — no
+            # fp32->bf16 permute/copy pass
+            dw = C.conv_wgrad_igemm(dy, x, R, R, ctx.stride, ctx.padding,
+                                    splits)
         return dx, dw, None, None
 
 

@@ -27,6 +27,8 @@ CONV_SHAPES = [
     (8, 64, 32, 32, 256, 1, 1, 0),     # resnet50 1x1 expand
     (8, 256, 8, 8, 1024, 1, 2, 0),     # resnet50 1x1 s2 shortcut
     (3, 64, 32, 32, 64, 3, 1, 1),      # ragged batch -> M % 128 != 0
+    (5, 3, 30, 30, 64, 3, 1, 1),       # ragged stem (PADC path, M % 64 != 0)
+    (8, 3, 224, 224, 64, 3, 1, 1),     # ImageNet-shape stem (PADC, big M)
 ]
 
 
@@ -82,9 +84,37 @@ def test_conv_wgrad_parity(shape):
     (y * dy).sum().backward()
     ref = w.grad
     K, C, R, _ = w.shape
-    dw = _C.conv_wgrad_igemm(dy.bfloat16(), x.bfloat16(), R, R, stride, pad)
-    got = dw.view(K, R, R, C).permute(0, 3, 1, 2)
+    # wgrad emits bf16 (K,C,R,S) channels_last directly
+    got = _C.conv_wgrad_igemm(dy.bfloat16(), x.bfloat16(), R, R, stride, pad)
+    assert got.shape == ref.shape
+    assert got.is_contiguous(memory_format=torch.channels_last)
     assert _rel_err(got, ref) < 2e-2, f"rel err {_rel_err(got, ref):.4f}"
+
+
+@pytest.mark.parametrize("splits", [1, 2, 32, 128])
+def test_conv_wgrad_splits_agree(splits):
+    """Every split-K factor reduces to the same weight grad (two-stage
+    slab reduce, no atomics) — and the stem's padded-channel path too."""
+    from mi355x_ddp import _C
+    for shape in [(8, 64, 32, 32, 64, 3, 1, 1), (8, 3, 32, 32, 64, 3, 1, 1)]:
+        x, w, stride, pad = _mk(shape)
+        y = F.conv2d(x, w, None, stride, pad)
+        dy = torch.randn_like(y).to(memory_format=torch.channels_last)
+        R = w.shape[2]
+        ref = _C.conv_wgrad_igemm(dy.bfloat16(), x.bfloat16(), R, R,
+                                  stride, pad, 0)
+        got = _C.conv_wgrad_igemm(dy.bfloat16(), x.bfloat16(), R, R,
+                                  stride, pad, splits)
+        assert _rel_err(got, ref.float()) < 5e-3
+
+
+def test_stem_fwd_tile_variants_agree():
+    """Stem PADC path: both GEMM-M tiles produce the same output."""
+    from mi355x_ddp import _C
+    x, w, stride, pad = _mk((8, 3, 32, 32, 64, 3, 1, 1))
+    a = _C.conv_fwd_igemm(x.bfloat16(), w.bfloat16(), stride, pad, 64)
+    b = _C.conv_fwd_igemm(x.bfloat16(), w.bfloat16(), stride, pad, 128)
+    assert torch.equal(a, b)
 
 
 def test_conv_autograd_function_end_to_end():

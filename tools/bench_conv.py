@@ -1,9 +1,13 @@
 #!/usr/bin/env python3
 """Per-shape micro-benchmark: native implicit-GEMM conv vs MIOpen (F.conv2d).
 
-Times fwd / dgrad / wgrad for every ResNet18 CIFAR conv shape at the bench
+Times fwd / dgrad / wgrad for every conv shape of a model config at the bench
 batch size, bf16 channels_last, and prints a table with speedups. Run on an
 MI355X box; output goes to stdout (redirect into gpurun_out/).
+
+Default: the ResNet18 CIFAR table (round-1 baseline format). --arch/--image-
+size collects the shape set from the actual model with forward hooks, e.g.
+`--arch resnet50 --image-size 224 --batch 64` for BASELINE config 5.
 """
 import argparse
 import os
@@ -13,6 +17,7 @@ import time
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
+import torch.nn as nn
 import torch.nn.functional as F
 
 
@@ -27,63 +32,109 @@ def timeit(fn, warmup=5, iters=20):
     return (time.perf_counter() - t0) / iters * 1e6  # us
 
 
+R18_CIFAR_SHAPES = [
+    # (name, C, H, W, K, R, stride, pad)
+    ("stem 3x3", 3, 32, 32, 64, 3, 1, 1),
+    ("s1 3x3", 64, 32, 32, 64, 3, 1, 1),
+    ("s2 3x3/2", 64, 32, 32, 128, 3, 2, 1),
+    ("s2 1x1/2", 64, 32, 32, 128, 1, 2, 0),
+    ("s2 3x3", 128, 16, 16, 128, 3, 1, 1),
+    ("s3 3x3/2", 128, 16, 16, 256, 3, 2, 1),
+    ("s3 3x3", 256, 8, 8, 256, 3, 1, 1),
+    ("s4 3x3/2", 256, 8, 8, 512, 3, 2, 1),
+    ("s4 3x3", 512, 4, 4, 512, 3, 1, 1),
+]
+
+
+def collect_model_shapes(arch: str, image_size: int):
+    """Unique conv shapes (with multiplicities) of a model at this input
+    size, discovered by a CPU forward pass with hooks."""
+    from mi355x_ddp.models import build_model
+    model = build_model(arch)
+    shapes = {}  # (C,H,W,K,R,stride,pad) -> count
+
+    def hook(mod, inp, out):
+        x = inp[0]
+        key = (x.shape[1], x.shape[2], x.shape[3], mod.out_channels,
+               mod.kernel_size[0], mod.stride[0], mod.padding[0])
+        shapes[key] = shapes.get(key, 0) + 1
+
+    handles = [m.register_forward_hook(hook) for m in model.modules()
+               if isinstance(m, nn.Conv2d)]
+    with torch.no_grad():
+        model(torch.zeros(1, 3, image_size, image_size))
+    for h in handles:
+        h.remove()
+    out = []
+    for (c, h, w, k, r, s, p), cnt in shapes.items():
+        tag = f"{r}x{r}" + (f"/{s}" if s > 1 else "")
+        out.append((f"{c}->{k} {tag}@{h}", c, h, w, k, r, s, p, cnt))
+    return out
+
+
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--batch", type=int, default=256)
+    p.add_argument("--arch", type=str, default=None,
+                   help="collect shapes from this model instead of the "
+                        "static ResNet18 CIFAR table")
+    p.add_argument("--image-size", type=int, default=32)
+    p.add_argument("--iters", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
     args = p.parse_args()
     from mi355x_ddp import _C
 
     N = args.batch
-    shapes = [
-        ("stem 3x3", N, 3, 32, 32, 64, 3, 1, 1),
-        ("s1 3x3", N, 64, 32, 32, 64, 3, 1, 1),
-        ("s2 3x3/2", N, 64, 32, 32, 128, 3, 2, 1),
-        ("s2 1x1/2", N, 64, 32, 32, 128, 1, 2, 0),
-        ("s2 3x3", N, 128, 16, 16, 128, 3, 1, 1),
-        ("s3 3x3/2", N, 128, 16, 16, 256, 3, 2, 1),
-        ("s3 3x3", N, 256, 8, 8, 256, 3, 1, 1),
-        ("s4 3x3/2", N, 256, 8, 8, 512, 3, 2, 1),
-        ("s4 3x3", N, 512, 4, 4, 512, 3, 1, 1),
-    ]
-    print(f"{'shape':>10} {'pass':>6} {'MIOpen us':>10} {'igemm us':>9} "
-          f"{'x':>6} {'TF/s':>7}")
-    tot_m = tot_i = 0.0
-    for name, n, c, h, w, k, r, stride, pad in shapes:
-        x = torch.randn(n, c, h, w, device="cuda", dtype=torch.bfloat16) \
+    if args.arch:
+        rows = collect_model_shapes(args.arch, args.image_size)
+    else:
+        rows = [(name, c, h, w, k, r, s, p, 1)
+                for (name, c, h, w, k, r, s, p) in R18_CIFAR_SHAPES]
+
+    print(f"{'shape':>18} {'pass':>6} {'n':>3} {'MIOpen us':>10} "
+          f"{'igemm us':>9} {'x':>6} {'TF/s':>7}")
+    tot_m = tot_i = 0.0  # weighted by multiplicity
+    for name, c, h, w, k, r, stride, pad, cnt in rows:
+        x = torch.randn(N, c, h, w, device="cuda", dtype=torch.bfloat16) \
             .to(memory_format=torch.channels_last)
         wt = (torch.randn(k, c, r, r, device="cuda", dtype=torch.bfloat16)
               / (c * r * r) ** 0.5).to(memory_format=torch.channels_last)
-        pq = ((h + 2 * pad - r) // stride + 1)
+        ph = ((h + 2 * pad - r) // stride + 1)
+        pw = ((w + 2 * pad - r) // stride + 1)
         y = F.conv2d(x, wt, None, stride, pad)
         dy = torch.randn_like(y).to(memory_format=torch.channels_last)
         wT = wt.flip(2, 3).permute(1, 2, 3, 0).contiguous()
-        flops = 2.0 * n * pq * pq * k * r * r * c
+        flops = 2.0 * N * ph * pw * k * r * r * c
 
+        def row(pss, tm, ti):
+            nonlocal tot_m, tot_i
+            tot_m += tm * cnt
+            tot_i += ti * cnt
+            print(f"{name:>18} {pss:>6} {cnt:>3} {tm:10.1f} {ti:9.1f} "
+                  f"{tm/ti:6.2f} {flops/ti/1e6:7.1f}", flush=True)
+
+        kw = dict(warmup=args.warmup, iters=args.iters)
         # fwd
-        tm = timeit(lambda: F.conv2d(x, wt, None, stride, pad))
-        ti = timeit(lambda: _C.conv_fwd_igemm(x, wt, stride, pad))
-        tot_m += tm; tot_i += ti
-        print(f"{name:>10} {'fwd':>6} {tm:10.1f} {ti:9.1f} {tm/ti:6.2f} "
-              f"{flops/ti/1e6:7.1f}")
+        tm = timeit(lambda: F.conv2d(x, wt, None, stride, pad), **kw)
+        ti = timeit(lambda: _C.conv_fwd_igemm(x, wt, stride, pad), **kw)
+        row("fwd", tm, ti)
         # dgrad (skip stem: input grad never needed there)
         if c != 3:
             xg = x.detach().requires_grad_(True)
             tm = timeit(lambda: torch.ops.aten.convolution_backward(
                 dy, xg, wt, None, [stride, stride], [pad, pad], [1, 1], False,
-                [0, 0], 1, [True, False, False]))
-            ti = timeit(lambda: _C.conv_dgrad_igemm(dy, wT, h, w, stride, pad))
-            tot_m += tm; tot_i += ti
-            print(f"{name:>10} {'dgrad':>6} {tm:10.1f} {ti:9.1f} {tm/ti:6.2f} "
-                  f"{flops/ti/1e6:7.1f}")
+                [0, 0], 1, [True, False, False]), **kw)
+            ti = timeit(lambda: _C.conv_dgrad_igemm(dy, wT, h, w, stride, pad),
+                        **kw)
+            row("dgrad", tm, ti)
         # wgrad
         tm = timeit(lambda: torch.ops.aten.convolution_backward(
             dy, x, wt, None, [stride, stride], [pad, pad], [1, 1], False,
-            [0, 0], 1, [False, True, False]))
-        ti = timeit(lambda: _C.conv_wgrad_igemm(dy, x, r, r, stride, pad))
-        tot_m += tm; tot_i += ti
-        print(f"{name:>10} {'wgrad':>6} {tm:10.1f} {ti:9.1f} {tm/ti:6.2f} "
-              f"{flops/ti/1e6:7.1f}")
-    print(f"{'TOTAL':>10} {'':>6} {tot_m:10.1f} {tot_i:9.1f} "
+            [0, 0], 1, [False, True, False]), **kw)
+        ti = timeit(lambda: _C.conv_wgrad_igemm(dy, x, r, r, stride, pad), **kw)
+        row("wgrad", tm, ti)
+        del x, wt, y, dy, wT
+    print(f"{'TOTAL':>18} {'':>6} {'':>3} {tot_m:10.1f} {tot_i:9.1f} "
           f"{tot_m/tot_i:6.2f}")
 
 
