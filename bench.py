@@ -48,6 +48,8 @@ def parse_args():
                    action="store_false")
     p.add_argument("--hip-graph", action="store_true")
     p.add_argument("--image-size", type=int, default=32)
+    p.add_argument("--comm-bf16", action="store_true",
+                   help="all-reduce gradients in bf16 (half the xGMI bytes)")
     return p.parse_args()
 
 
@@ -79,6 +81,7 @@ def main():
         sync_bn=(not args.no_syncbn) and world > 1,
         grad_accu_steps=args.grad_accu_steps,
         channels_last=args.channels_last, hip_graph=args.hip_graph,
+        comm_bf16=args.comm_bf16,
         use_flat_ddp=(args.mode == "flat"), synthetic=True)
     per_rank = cfg.per_rank_batch(world)
 

@@ -1,13 +1,21 @@
-"""hipGraph-captured training step.
+"""hipGraph-captured training step (opt-in, `--hip-graph`).
 
-The ResNet18/CIFAR step is launch-bound on MI355X (~100+ small kernels over a
-few ms), so the whole step — zero-grad, forward, loss, backward, flat gradient
+The whole step — zero-grad, forward, loss, backward, flat gradient
 all-reduce, fused SGD — is captured once into a hipGraph and replayed per
 batch; per-step host work collapses to two H2D copies + one graph launch.
-Requires the FlatDDP wrapper (static grad memory: param.grad are views of one
-flat buffer) and the arithmetic-count BN path (no host syncs inside the step).
-RCCL collectives are capturable, so the world_size>1 all-reduce is inside the
-graph too.
+Requires FlatDDP with static_grads=True (param.grad are views of one flat
+buffer) and the arithmetic-count BN path (no host syncs inside the step).
+RCCL collectives are capturable, so the world_size>1 all-reduce is inside
+the graph too.
+
+Measured honestly (round 1, 1x MI355X, ResNet18/CIFAR bf16): graph replay is
+~3-5% SLOWER than the eager step at world 1 — the eager launch stream already
+overlaps with GPU work at this kernel size, while replay serialises the H2D
+static-buffer copies with the replay and pays the ~10-16 us replay floor
+(profiles/bench_results_r1.md: 35.2-36.9k img/s graphed vs 38.7-47.2k eager).
+It exists because it is the launch-overhead-proof variant: the win shows up
+when per-step host work grows (small models at very small batch, or a
+host-jittery rank fleet), not on this benchmark. Default stays eager.
 """
 from __future__ import annotations
 

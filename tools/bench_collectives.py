@@ -30,30 +30,37 @@ def main():
     dev = torch.device("cuda", local) if torch.cuda.is_available() else "cpu"
 
     sizes = [4, 1024, 64 * 1024, 256 * 1024, 1 << 20, 4 << 20, 12 << 20,
-             25 << 20, 45 << 20, 64 << 20]  # bytes (fp32 elements / 4)
+             25 << 20, 45 << 20, 64 << 20]  # bytes
     if rank == 0:
-        print(f"world={world}  backend={backend}")
-        print(f"{'bytes':>12} {'us/call':>10} {'algbw GB/s':>11} "
+        print(f"world={world}  backend={backend}  "
+              f"NCCL_ALGO={os.environ.get('NCCL_ALGO', '<default>')}")
+        print(f"{'dtype':>6} {'bytes':>12} {'us/call':>10} {'algbw GB/s':>11} "
               f"{'busbw GB/s':>11}")
-    for nbytes in sizes:
-        t = torch.ones(nbytes // 4, device=dev)
-        for _ in range(5):
-            dist.all_reduce(t)
-        if dev != "cpu":
-            torch.cuda.synchronize()
-        dist.barrier()
-        iters = 20
-        t0 = time.perf_counter()
-        for _ in range(iters):
-            dist.all_reduce(t)
-        if dev != "cpu":
-            torch.cuda.synchronize()
-        el = (time.perf_counter() - t0) / iters
-        if rank == 0:
-            algbw = nbytes / el / 1e9
-            busbw = algbw * 2 * (world - 1) / max(1, world)
-            print(f"{nbytes:>12} {el * 1e6:>10.1f} {algbw:>11.2f} "
-                  f"{busbw:>11.2f}", flush=True)
+    # fp32 = FlatDDP's default gradient buckets; bf16 = comm_dtype=bf16
+    # shadow buckets (same element counts, half the wire bytes)
+    for dtype, esz in ((torch.float32, 4), (torch.bfloat16, 2)):
+        if dtype == torch.bfloat16 and backend == "gloo" and dev == "cpu":
+            continue  # CPU-gloo bf16 all_reduce support varies; fp32 suffices
+        for nbytes in sizes:
+            t = torch.ones(nbytes // esz, device=dev, dtype=dtype)
+            for _ in range(5):
+                dist.all_reduce(t)
+            if dev != "cpu":
+                torch.cuda.synchronize()
+            dist.barrier()
+            iters = 20
+            t0 = time.perf_counter()
+            for _ in range(iters):
+                dist.all_reduce(t)
+            if dev != "cpu":
+                torch.cuda.synchronize()
+            el = (time.perf_counter() - t0) / iters
+            if rank == 0:
+                algbw = nbytes / el / 1e9
+                busbw = algbw * 2 * (world - 1) / max(1, world)
+                name = "fp32" if esz == 4 else "bf16"
+                print(f"{name:>6} {nbytes:>12} {el * 1e6:>10.1f} "
+                      f"{algbw:>11.2f} {busbw:>11.2f}", flush=True)
     dist.destroy_process_group()
 
 
