@@ -504,6 +504,189 @@ void conv_wgrad_kernel(const __bf16* __restrict__ dy,
     }
 }
 
+// ---------------------------------------------------------------------------
+// wgrad v2: large-tile variant for the big-spatial / big-channel shapes
+// (ResNet50 @ 224px). Tile = (WGM*64) x (WGN*64) per 256-thread block
+// ((2,2) -> 128x128, (4,1) -> 256x64): fewer K-passes over dy/x than v1's
+// 64x128 (cross-tile re-reads scale with tile count), and the LDS transpose
+// is done 4x4 in REGISTERS (8 v_perm per block) with ds_write_b64 stores —
+// v1 staged with 48 scalar b16 LDS writes per thread per chunk, which
+// rivalled the MFMA issue time. FAST-path shapes only (C % 64 == 0).
+// ---------------------------------------------------------------------------
+
+// transpose a 4x4 bf16 block: in[i] = row i as uint2 {e[i][0..1], e[i][2..3]},
+// out[c] = column c. 8 v_perm_b32: selector picks [b0,b1,a0,a1]/[b2,b3,a2,a3]
+// bytes of the {src1(b) low, src0(a) high} concatenation.
+__device__ __forceinline__ void tr4x4_bf16(const uint2 in[4], uint2 out[4]) {
+  const unsigned LO = 0x05040100u, HI = 0x07060302u;
+  out[0].x = __builtin_amdgcn_perm(in[1].x, in[0].x, LO);
+  out[0].y = __builtin_amdgcn_perm(in[3].x, in[2].x, LO);
+  out[1].x = __builtin_amdgcn_perm(in[1].x, in[0].x, HI);
+  out[1].y = __builtin_amdgcn_perm(in[3].x, in[2].x, HI);
+  out[2].x = __builtin_amdgcn_perm(in[1].y, in[0].y, LO);
+  out[2].y = __builtin_amdgcn_perm(in[3].y, in[2].y, LO);
+  out[3].x = __builtin_amdgcn_perm(in[1].y, in[0].y, HI);
+  out[3].y = __builtin_amdgcn_perm(in[3].y, in[2].y, HI);
+}
+
+template <int WGM, int WGN, bool DIRECT>
+__global__ __launch_bounds__(256)
+void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
+                          const __bf16* __restrict__ x,
+                          float* __restrict__ dwp,
+                          __bf16* __restrict__ dwb, ConvDims d) {
+  constexpr int TM = WGM * 64;          // kout tile
+  constexpr int TN = WGN * 64;          // rsc tile
+  __shared__ __bf16 sA[TM * LDK];       // [kout][npq]
+  __shared__ __bf16 sB[TN * LDK];       // [rsc][npq]
+  // per-chunk row metadata: for npq row r, the x base offset of tap (0,0)
+  // and the (ih0, iw0) coords for bounds tests (computed once per chunk by
+  // threads 0..63 instead of 4x per staged 4x4 block)
+  __shared__ int sRowOff[64];           // ((xn*GH + ih0)*GW + iw0) or -1
+  __shared__ short sIh0[64], sIw0[64];
+
+  const int tid = threadIdx.x;
+  const int m0 = blockIdx.y * TM;
+  const int n0 = blockIdx.x * TN;
+
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wm = (wave / WGN) * 64;
+  const int wn = (wave % WGN) * 64;
+  const int fr = lane & 15;
+  const int fk = (lane >> 4) * 8;
+
+  const int SC = d.S * d.GC;
+  const int nchunks = (d.K + BK - 1) / BK;
+  f32x4 acc[4][4] = {};
+
+  for (int kc = blockIdx.z; kc < nchunks; kc += gridDim.z) {
+    const int kk0 = kc * BK;
+    // ---- row metadata (threads 0..63) --------------------------------
+    if (tid < 64) {
+      const int npq = kk0 + tid;
+      int off = -1, ih0 = 0, iw0 = 0;
+      if (npq < d.K) {
+        int xn, xp, xq;
+        decode_m(npq, d, xn, xp, xq);
+        ih0 = xp * d.stride - d.pad;
+        iw0 = xq * d.stride - d.pad;
+        off = ((xn * d.GH + ih0) * d.GW + iw0);
+      }
+      sRowOff[tid] = off;
+      sIh0[tid] = (short)ih0;
+      sIw0[tid] = (short)iw0;
+    }
+    __syncthreads();
+
+    // ---- dy tile: TM x 64, 4x4-register-transposed -------------------
+    {
+      constexpr int TA = TM / 64;       // 4x4 blocks per thread
+      #pragma unroll
+      for (int t = 0; t < TA; ++t) {
+        const int bid = tid * TA + t;
+        const int c4 = bid >> 4;        // kout 4-col group
+        const int r4 = bid & 15;        // npq 4-row group
+        const int npq0 = kk0 + r4 * 4;
+        const int mcol = m0 + c4 * 4;
+        uint2 in[4], out[4];
+        if (npq0 + 4 <= d.K && mcol + 4 <= d.M) {
+          #pragma unroll
+          for (int i = 0; i < 4; ++i)
+            in[i] = *(const uint2*)(dy + (long)(npq0 + i) * d.M + mcol);
+        } else {
+          #pragma unroll
+          for (int i = 0; i < 4; ++i) {
+            __bf16 e[4] = {};
+            if (npq0 + i < d.K)
+              #pragma unroll
+              for (int c = 0; c < 4; ++c)
+                if (mcol + c < d.M)
+                  e[c] = dy[(long)(npq0 + i) * d.M + mcol + c];
+            in[i] = *(uint2*)e;
+          }
+        }
+        tr4x4_bf16(in, out);
+        #pragma unroll
+        for (int i = 0; i < 4; ++i)
+          *(uint2*)(sA + (c4 * 4 + i) * LDK + r4 * 4) = out[i];
+      }
+    }
+    // ---- x tile: TN x 64, gather via row metadata --------------------
+    {
+      constexpr int TB = TN / 64;
+      #pragma unroll
+      for (int t = 0; t < TB; ++t) {
+        const int bid = tid * TB + t;
+        const int c4 = bid >> 4;
+        const int r4 = bid & 15;
+        const int nn = n0 + c4 * 4;     // rsc col of this 4-col group
+        // FAST contract (C % 64 == 0): nn..nn+3 sit inside ONE filter tap
+        const int i_tap = nn / SC;
+        const int rem = nn - i_tap * SC;
+        const int j_tap = rem / d.GC;
+        const int cch = rem - j_tap * d.GC;
+        uint2 in[4], out[4];
+        const bool ncol_ok = nn + 4 <= d.N;
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int r = r4 * 4 + i;
+          uint2 v = {0u, 0u};
+          if (ncol_ok && sRowOff[r] >= 0) {
+            const int ih = sIh0[r] + i_tap;
+            const int iw = sIw0[r] + j_tap;
+            if (ih >= 0 && ih < d.GH && iw >= 0 && iw < d.GW)
+              v = *(const uint2*)(x +
+                  ((long)sRowOff[r] + i_tap * d.GW + j_tap) * d.GC + cch);
+          }
+          in[i] = v;
+        }
+        tr4x4_bf16(in, out);
+        #pragma unroll
+        for (int i = 0; i < 4; ++i)
+          *(uint2*)(sB + (c4 * 4 + i) * LDK + r4 * 4) = out[i];
+      }
+    }
+    __syncthreads();
+
+    #pragma unroll
+    for (int ks = 0; ks < BK; ks += 32) {
+      bf16x8 af[4], bf[4];
+      #pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        af[mi] = *(const bf16x8*)&sA[(wm + mi * 16 + fr) * LDK + ks + fk];
+      #pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        bf[ni] = *(const bf16x8*)&sB[(wn + ni * 16 + fr) * LDK + ks + fk];
+      #pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int dm = (lane >> 4) * 4;
+  const int dn = lane & 15;
+  float* slab = DIRECT ? nullptr : dwp + (long)blockIdx.z * d.M * d.N;
+  #pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+    #pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      const int n = n0 + wn + ni * 16 + dn;
+      if (n >= d.N) continue;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = m0 + wm + mi * 16 + dm + r;
+        if (m >= d.M) continue;
+        if (DIRECT) dwb[(long)m * d.N + n] = (__bf16)acc[mi][ni][r];
+        else slab[(long)m * d.N + n] = acc[mi][ni][r];
+      }
+    }
+}
+
 // Sum the split-K partial slabs and emit the bf16 channels_last weight grad:
 // out[k][tap*Cout + c] (= memory layout of channels_last (K, Cout, R, S)).
 // Cpad==Cout -> identity column map; the stem maps Cpad=4 -> Cout=3 and drops
@@ -742,8 +925,11 @@ at::Tensor conv_dgrad_igemm(at::Tensor dy, at::Tensor wT, long H, long W,
 // dw bf16 (K,C,R,S) channels_last <- dy, x. Two-stage split-K: private fp32
 // partial slabs per grid.z slice + one reduce pass (no atomics); splits==1
 // non-stem collapses to a DIRECT bf16 store. `splits`: 0 = heuristic.
+// `wtile`: 0 = shape heuristic, 1 = v1 64x128, 2 = v2 128x128, 3 = v2 256x64
+// (the autotune cache measures and pins these per shape).
 at::Tensor conv_wgrad_igemm(at::Tensor dy, at::Tensor x, long R, long S,
-                            long stride, long pad, long splits_arg) {
+                            long stride, long pad, long splits_arg,
+                            long wtile) {
   check_nhwc_bf16(dy, "dy"); check_nhwc_bf16(x, "x");
   const int Nb = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   const int K = dy.size(1), P = dy.size(2), Q = dy.size(3);
@@ -769,19 +955,36 @@ at::Tensor conv_wgrad_igemm(at::Tensor dy, at::Tensor x, long R, long S,
 
   ConvDims d{Nb, P, Q, H, W, Cpad, (int)R, (int)S, (int)stride, (int)pad,
              K, Npad, Nb * P * Q};
-  const int tm = (d.M + 63) / 64, tn = (d.N + 127) / 128;
+  const bool fast = (C % BK == 0) && (K % 16 == 0);
+
+  // tile pick: v2's larger tiles read dy/x fewer times (traffic per pass
+  // scales with the tile count along the other axis) and stage with wide
+  // LDS writes; v1 remains for small-M shapes and the generic/stem paths
+  int tile = (int)wtile;
+  if (tile == 0) {
+    if (fast && !stem && d.M >= 256 && d.N <= 64) tile = 3;
+    else if (fast && !stem && d.M >= 128 && d.N >= 96) tile = 2;
+    else tile = 1;
+  }
+  if (!fast || stem) tile = 1;
+  const int TM = tile == 3 ? 256 : (tile == 2 ? 128 : 64);
+  const int TN = tile == 1 ? 128 : (tile == 2 ? 128 : 64);
+
+  const int tm = (d.M + TM - 1) / TM, tn = (d.N + TN - 1) / TN;
   const int nchunks = (d.K + BK - 1) / BK;
-  int splits = (splits_arg > 0) ? (int)splits_arg : 768 / (tm * tn);
+  int splits = (splits_arg > 0) ? (int)splits_arg : 1024 / (tm * tn);
   // bound the partial-slab workspace to ~96 MB
   const long max_ws = 96L * 1024 * 1024 / ((long)d.M * d.N * 4);
   splits = std::max(1, (int)std::min({(long)splits, (long)nchunks,
-                                      std::max(max_ws, 1L), 256L}));
+                                      std::max(max_ws, 1L), 1024L}));
   const dim3 grid(tn, tm, splits);
-  const bool fast = (C % BK == 0) && (K % 16 == 0);
 
   if (splits == 1 && !stem) {
-    auto* kern = fast ? conv_wgrad_kernel<true, false, true>
-                      : conv_wgrad_kernel<false, false, true>;
+    auto* kern =
+        tile == 3 ? conv_wgrad_v2_kernel<4, 1, true>
+        : tile == 2 ? conv_wgrad_v2_kernel<2, 2, true>
+        : (fast ? conv_wgrad_kernel<true, false, true>
+                : conv_wgrad_kernel<false, false, true>);
     hipLaunchKernelGGL(kern, grid, dim3(256), 0, stream, bf16_ptr(dy), xp,
                        nullptr, reinterpret_cast<__bf16*>(dw.data_ptr()), d);
     return dw;
@@ -789,9 +992,12 @@ at::Tensor conv_wgrad_igemm(at::Tensor dy, at::Tensor x, long R, long S,
 
   auto part = at::empty({(long)splits * d.M * d.N},
                         x.options().dtype(at::kFloat));
-  auto* kern = stem ? conv_wgrad_kernel<false, true, false>
-                    : (fast ? conv_wgrad_kernel<true, false, false>
-                            : conv_wgrad_kernel<false, false, false>);
+  auto* kern =
+      tile == 3 ? conv_wgrad_v2_kernel<4, 1, false>
+      : tile == 2 ? conv_wgrad_v2_kernel<2, 2, false>
+      : stem ? conv_wgrad_kernel<false, true, false>
+             : (fast ? conv_wgrad_kernel<true, false, false>
+                     : conv_wgrad_kernel<false, false, false>);
   hipLaunchKernelGGL(kern, grid, dim3(256), 0, stream, bf16_ptr(dy), xp,
                      part.data_ptr<float>(), nullptr, d);
   const long MN = (long)d.M * d.N;

@@ -1288,7 +1288,7 @@ at::Tensor conv_fwd_igemm(at::Tensor x, at::Tensor w, long stride, long pad,
 at::Tensor conv_dgrad_igemm(at::Tensor dy, at::Tensor wT, long H, long W,
                             long stride, long pad, long tile);
 at::Tensor conv_wgrad_igemm(at::Tensor dy, at::Tensor x, long R, long S,
-                            long stride, long pad, long splits);
+                            long stride, long pad, long splits, long wtile);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_build_wT", &conv_build_wT,
@@ -1303,9 +1303,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("stride"), py::arg("pad"), py::arg("tile") = 0);
   m.def("conv_wgrad_igemm", &conv_wgrad_igemm,
         "implicit-GEMM conv weight-grad -> bf16 channels_last (K,C,R,S); "
-        "two-stage split-K, no atomics; splits 0=auto",
+        "two-stage split-K, no atomics; splits 0=auto; wtile 0=auto, "
+        "1=64x128, 2=128x128, 3=256x64",
         py::arg("dy"), py::arg("x"), py::arg("R"), py::arg("S"),
-        py::arg("stride"), py::arg("pad"), py::arg("splits") = 0);
+        py::arg("stride"), py::arg("pad"), py::arg("splits") = 0,
+        py::arg("wtile") = 0);
   m.def("bn_stats", &bn_stats, "per-channel sum/sqsum (NCHW)");
   m.def("bn_stats_packed", &bn_stats_packed,
         "per-channel {sum,sqsum} packed [2C], no-atomic NHWC v2");

@@ -72,14 +72,19 @@ def _measure_ms(fn: Callable[[], None], iters: int = 4) -> float:
     return start.elapsed_time(stop)
 
 
-def _tuned_choice(key: Tuple, candidates: Sequence[int],
-                  run: Callable[[int], None]) -> int:
-    """Cached fastest candidate for this shape key (0 = kernel heuristic)."""
+def _tuned_choice(key: Tuple, candidates: Sequence,
+                  run: Callable, default=None):
+    """Cached fastest candidate for this shape key.
+
+    `default` (first candidate unless given) is returned without measuring
+    when autotune is off or a hipGraph capture is in flight — for the native
+    kernels that is the launch-config heuristic, never MIOpen.
+    """
     if key in _TUNE_CACHE:
         return _TUNE_CACHE[key]
     if not autotune_wanted() or torch.cuda.is_current_stream_capturing():
-        return 0
-    best, best_ms = 0, float("inf")
+        return candidates[0] if default is None else default
+    best, best_ms = candidates[0], float("inf")
     for c in candidates:
         ms = _measure_ms(lambda: run(c))
         if ms < best_ms:
@@ -92,7 +97,14 @@ def clear_autotune_cache() -> None:
     _TUNE_CACHE.clear()
 
 
+MIOPEN = -1  # tuner sentinel: this pass is fastest on the MIOpen kernel
+
+
 class _ConvIGEMM(torch.autograd.Function):
+    """Per-pass dispatch between the native igemm kernels (with their tile /
+    split-K launch knobs) and MIOpen, chosen by the per-shape autotune cache.
+    The heuristic default (autotune off) is always the native kernel."""
+
     @staticmethod
     def forward(ctx, x, weight, stride: int, padding: int):
         x = x.contiguous(memory_format=torch.channels_last)
@@ -100,11 +112,14 @@ class _ConvIGEMM(torch.autograd.Function):
         ctx.save_for_backward(x, weight)
         ctx.stride, ctx.padding = stride, padding
         C = _backend.C()
+
+        def run(t):
+            if t == MIOPEN:
+                return F.conv2d(x, weight, None, stride, padding)
+            return C.conv_fwd_igemm(x, weight, stride, padding, t)
+
         key = ("fwd", x.shape, weight.shape, stride, padding)
-        tile = _tuned_choice(
-            key, (64, 128),
-            lambda t: C.conv_fwd_igemm(x, weight, stride, padding, t))
-        return C.conv_fwd_igemm(x, weight, stride, padding, tile)
+        return run(_tuned_choice(key, (64, 128, MIOPEN), run, default=0))
 
     @staticmethod
     def backward(ctx, dy):
@@ -115,24 +130,49 @@ class _ConvIGEMM(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             # 180°-rotated, (C,R,S,K)-transposed filter for the dgrad GEMM
             wT = C.conv_build_wT(weight)
+
+            def run_dx(t):
+                if t == MIOPEN:
+                    return torch.ops.aten.convolution_backward(
+                        dy, x, weight, None,
+                        [ctx.stride, ctx.stride], [ctx.padding, ctx.padding],
+                        [1, 1], False, [0, 0], 1, [True, False, False])[0]
+                return C.conv_dgrad_igemm(dy, wT, x.shape[2], x.shape[3],
+                                          ctx.stride, ctx.padding, t)
+
             key = ("dgrad", dy.shape, wT.shape, ctx.stride, ctx.padding)
-            tile = _tuned_choice(
-                key, (64, 128),
-                lambda t: C.conv_dgrad_igemm(dy, wT, x.shape[2], x.shape[3],
-                                             ctx.stride, ctx.padding, t))
-            dx = C.conv_dgrad_igemm(dy, wT, x.shape[2], x.shape[3],
-                                    ctx.stride, ctx.padding, tile)
+            dx = run_dx(_tuned_choice(key, (64, 128, MIOPEN), run_dx,
+                                      default=0))
         if ctx.needs_input_grad[1]:
             R = weight.shape[2]
-            key = ("wgrad", dy.shape, x.shape, R, ctx.stride, ctx.padding)
-            splits = _tuned_choice(
-                key, (0, 1, 32, 128),
-                lambda s: C.conv_wgrad_igemm(dy, x, R, R, ctx.stride,
-                                             ctx.padding, s))
-            # kernel emits bf16 (K,C,R,S) channels_last directly — no
-            # fp32->bf16 permute/copy pass
-            dw = C.conv_wgrad_igemm(dy, x, R, R, ctx.stride, ctx.padding,
-                                    splits)
+
+            def run_dw(c):
+                if c == MIOPEN:
+                    g = torch.ops.aten.convolution_backward(
+                        dy, x, weight, None,
+                        [ctx.stride, ctx.stride], [ctx.padding, ctx.padding],
+                        [1, 1], False, [0, 0], 1, [False, True, False])[1]
+                    return g.contiguous(memory_format=torch.channels_last)
+                wtile, splits = c
+                # kernel emits bf16 (K,C,R,S) channels_last directly — no
+                # fp32->bf16 permute/copy pass
+                return C.conv_wgrad_igemm(dy, x, R, R, ctx.stride,
+                                          ctx.padding, splits, wtile)
+
+            # two-phase: pick the tile (v1/v2 variants vs MIOpen), then the
+            # split-K factor for the winning native tile
+            kt = ("wgrad_tile", dy.shape, x.shape, R, ctx.stride, ctx.padding)
+            tile = _tuned_choice(kt, ((0, 0), (1, 0), (2, 0), (3, 0), MIOPEN),
+                                 run_dw, default=(0, 0))
+            if tile == MIOPEN:
+                choice = MIOPEN
+            else:
+                ks = ("wgrad_splits", dy.shape, x.shape, R, ctx.stride,
+                      ctx.padding, tile[0])
+                choice = _tuned_choice(
+                    ks, tuple((tile[0], s) for s in (0, 1, 32, 128)), run_dw,
+                    default=tile)
+            dw = run_dw(choice)
         return dx, dw, None, None
 
 

@@ -108,6 +108,28 @@ def test_conv_wgrad_splits_agree(splits):
         assert _rel_err(got, ref.float()) < 5e-3
 
 
+@pytest.mark.parametrize("wtile", [1, 2, 3])
+def test_conv_wgrad_tile_variants_agree(wtile):
+    """v2's 128x128 / 256x64 tiles (register 4x4 transpose + ds_write_b64
+    staging) compute the same dw as the v1 64x128 tile."""
+    from mi355x_ddp import _C
+    shapes = [(8, 64, 32, 32, 64, 3, 1, 1),     # M=64 (v2 falls back bounds)
+              (8, 64, 32, 32, 256, 1, 1, 0),    # M=256, N=64 (256x64 target)
+              (8, 128, 16, 16, 128, 3, 1, 1),   # M=128, N=1152 (128x128)
+              (8, 256, 8, 8, 512, 3, 2, 1)]
+    for shape in shapes:
+        x, w, stride, pad = _mk(shape)
+        y = F.conv2d(x, w, None, stride, pad)
+        dy = torch.randn_like(y).to(memory_format=torch.channels_last)
+        R = w.shape[2]
+        ref = _C.conv_wgrad_igemm(dy.bfloat16(), x.bfloat16(), R, R,
+                                  stride, pad, 1, 1)  # v1, single split
+        got = _C.conv_wgrad_igemm(dy.bfloat16(), x.bfloat16(), R, R,
+                                  stride, pad, 0, wtile)
+        assert _rel_err(got, ref.float()) < 5e-3, \
+            f"{shape} wtile={wtile}: {_rel_err(got, ref.float()):.4f}"
+
+
 def test_stem_fwd_tile_variants_agree():
     """Stem PADC path: both GEMM-M tiles produce the same output."""
     from mi355x_ddp import _C

@@ -92,8 +92,8 @@ def main():
                 for (name, c, h, w, k, r, s, p) in R18_CIFAR_SHAPES]
 
     print(f"{'shape':>18} {'pass':>6} {'n':>3} {'MIOpen us':>10} "
-          f"{'igemm us':>9} {'x':>6} {'TF/s':>7}")
-    tot_m = tot_i = 0.0  # weighted by multiplicity
+          f"{'igemm us':>9} {'best us':>9} {'cfg':>9} {'x':>6} {'TF/s':>7}")
+    tot_m = tot_i = tot_b = 0.0  # weighted by multiplicity
     for name, c, h, w, k, r, stride, pad, cnt in rows:
         x = torch.randn(N, c, h, w, device="cuda", dtype=torch.bfloat16) \
             .to(memory_format=torch.channels_last)
@@ -106,18 +106,30 @@ def main():
         wT = wt.flip(2, 3).permute(1, 2, 3, 0).contiguous()
         flops = 2.0 * N * ph * pw * k * r * r * c
 
-        def row(pss, tm, ti):
-            nonlocal tot_m, tot_i
+        def row(pss, tm, ti, tb, cfg):
+            nonlocal tot_m, tot_i, tot_b
             tot_m += tm * cnt
             tot_i += ti * cnt
+            tot_b += tb * cnt
             print(f"{name:>18} {pss:>6} {cnt:>3} {tm:10.1f} {ti:9.1f} "
-                  f"{tm/ti:6.2f} {flops/ti/1e6:7.1f}", flush=True)
+                  f"{tb:9.1f} {cfg:>9} {tm/tb:6.2f} {flops/tb/1e6:7.1f}",
+                  flush=True)
+
+        def sweep(cands, run):
+            best, bcfg = float("inf"), None
+            for cand in cands:
+                t = timeit(lambda: run(cand), warmup=2, iters=args.iters)
+                if t < best:
+                    best, bcfg = t, cand
+            return best, str(bcfg)
 
         kw = dict(warmup=args.warmup, iters=args.iters)
         # fwd
         tm = timeit(lambda: F.conv2d(x, wt, None, stride, pad), **kw)
         ti = timeit(lambda: _C.conv_fwd_igemm(x, wt, stride, pad), **kw)
-        row("fwd", tm, ti)
+        tb, cfg = sweep((64, 128),
+                        lambda t: _C.conv_fwd_igemm(x, wt, stride, pad, t))
+        row("fwd", tm, ti, tb, cfg)
         # dgrad (skip stem: input grad never needed there)
         if c != 3:
             xg = x.detach().requires_grad_(True)
@@ -126,16 +138,22 @@ def main():
                 [0, 0], 1, [True, False, False]), **kw)
             ti = timeit(lambda: _C.conv_dgrad_igemm(dy, wT, h, w, stride, pad),
                         **kw)
-            row("dgrad", tm, ti)
-        # wgrad
+            tb, cfg = sweep((64, 128), lambda t: _C.conv_dgrad_igemm(
+                dy, wT, h, w, stride, pad, t))
+            row("dgrad", tm, ti, tb, cfg)
+        # wgrad: sweep tile variants x split factors
         tm = timeit(lambda: torch.ops.aten.convolution_backward(
             dy, x, wt, None, [stride, stride], [pad, pad], [1, 1], False,
             [0, 0], 1, [False, True, False]), **kw)
         ti = timeit(lambda: _C.conv_wgrad_igemm(dy, x, r, r, stride, pad), **kw)
-        row("wgrad", tm, ti)
+        wcands = [(t, s) for t in (1, 2, 3) for s in (0, 1, 32, 128)] \
+            if c != 3 else [(1, s) for s in (0, 32, 128, 512)]
+        tb, cfg = sweep(wcands, lambda ts: _C.conv_wgrad_igemm(
+            dy, x, r, r, stride, pad, ts[1], ts[0]))
+        row("wgrad", tm, ti, tb, cfg)
         del x, wt, y, dy, wT
     print(f"{'TOTAL':>18} {'':>6} {'':>3} {tot_m:10.1f} {tot_i:9.1f} "
-          f"{tot_m/tot_i:6.2f}")
+          f"{tot_b:9.1f} {'':>9} {tot_m/tot_b:6.2f}")
 
 
 if __name__ == "__main__":
