@@ -346,7 +346,11 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
 // contiguous per row) so MFMA fragment reads are 16-byte ds_read_b128.
 // PADC: x has 4 padded channels (stem) — per-tap dwordx2 gathers.
 // ---------------------------------------------------------------------------
-template <bool FAST, bool PADC = false, bool DIRECT = false>
+// ATOMIC: all grid.z slices atomicAdd into ONE fp32 slab (round 1's scheme)
+// instead of private slabs — wins on small dw where the slab+reduce pass
+// costs more than the contention; kept as autotune variant wtile=4.
+template <bool FAST, bool PADC = false, bool DIRECT = false,
+          bool ATOMIC = false>
 __global__ __launch_bounds__(256)
 void conv_wgrad_kernel(const __bf16* __restrict__ dy,
                        const __bf16* __restrict__ x,
@@ -487,7 +491,8 @@ void conv_wgrad_kernel(const __bf16* __restrict__ dy,
 
   const int dm = (lane >> 4) * 4;
   const int dn = lane & 15;
-  float* slab = DIRECT ? nullptr : dwp + (long)blockIdx.z * d.M * d.N;
+  float* slab = DIRECT ? nullptr
+              : dwp + (ATOMIC ? 0L : (long)blockIdx.z * d.M * d.N);
   #pragma unroll
   for (int mi = 0; mi < 2; ++mi)
     #pragma unroll
@@ -499,6 +504,7 @@ void conv_wgrad_kernel(const __bf16* __restrict__ dy,
         const int m = m0 + wm + mi * 16 + dm + r;
         if (m >= d.M) continue;
         if (DIRECT) dwb[(long)m * d.N + n] = (__bf16)acc[mi][ni][r];
+        else if (ATOMIC) atomicAdd(&slab[(long)m * d.N + n], acc[mi][ni][r]);
         else slab[(long)m * d.N + n] = acc[mi][ni][r];
       }
     }
@@ -966,20 +972,24 @@ at::Tensor conv_wgrad_igemm(at::Tensor dy, at::Tensor x, long R, long S,
     else if (fast && !stem && d.M >= 128 && d.N >= 96) tile = 2;
     else tile = 1;
   }
-  if (!fast || stem) tile = 1;
+  if (!fast || stem) tile = (tile == 4) ? 4 : 1;  // v2 needs the FAST layout
   const int TM = tile == 3 ? 256 : (tile == 2 ? 128 : 64);
-  const int TN = tile == 1 ? 128 : (tile == 2 ? 128 : 64);
+  const int TN = tile == 2 ? 128 : (tile == 3 ? 64 : 128);
 
   const int tm = (d.M + TM - 1) / TM, tn = (d.N + TN - 1) / TN;
   const int nchunks = (d.K + BK - 1) / BK;
-  int splits = (splits_arg > 0) ? (int)splits_arg : 1024 / (tm * tn);
+  // heuristic: ~1024 blocks fills the chip, but keep >=8 chunks per block
+  // (fewer and the per-block fill/drain + slab traffic dominates — measured
+  // on the CIFAR shapes, where over-splitting cost 2-4x)
+  int splits = (splits_arg > 0) ? (int)splits_arg
+                                : std::min(1024 / (tm * tn), nchunks / 8);
   // bound the partial-slab workspace to ~96 MB
   const long max_ws = 96L * 1024 * 1024 / ((long)d.M * d.N * 4);
   splits = std::max(1, (int)std::min({(long)splits, (long)nchunks,
                                       std::max(max_ws, 1L), 1024L}));
   const dim3 grid(tn, tm, splits);
 
-  if (splits == 1 && !stem) {
+  if (splits == 1 && !stem && tile != 4) {
     auto* kern =
         tile == 3 ? conv_wgrad_v2_kernel<4, 1, true>
         : tile == 2 ? conv_wgrad_v2_kernel<2, 2, true>
@@ -990,21 +1000,26 @@ at::Tensor conv_wgrad_igemm(at::Tensor dy, at::Tensor x, long R, long S,
     return dw;
   }
 
-  auto part = at::empty({(long)splits * d.M * d.N},
-                        x.options().dtype(at::kFloat));
+  const bool atomic = (tile == 4);
+  const long MN = (long)d.M * d.N;
+  auto part = atomic
+      ? at::zeros({MN}, x.options().dtype(at::kFloat))
+      : at::empty({(long)splits * MN}, x.options().dtype(at::kFloat));
   auto* kern =
       tile == 3 ? conv_wgrad_v2_kernel<4, 1, false>
       : tile == 2 ? conv_wgrad_v2_kernel<2, 2, false>
+      : atomic ? (stem ? conv_wgrad_kernel<false, true, false, true>
+                 : fast ? conv_wgrad_kernel<true, false, false, true>
+                        : conv_wgrad_kernel<false, false, false, true>)
       : stem ? conv_wgrad_kernel<false, true, false>
              : (fast ? conv_wgrad_kernel<true, false, false>
                      : conv_wgrad_kernel<false, false, false>);
   hipLaunchKernelGGL(kern, grid, dim3(256), 0, stream, bf16_ptr(dy), xp,
                      part.data_ptr<float>(), nullptr, d);
-  const long MN = (long)d.M * d.N;
   hipLaunchKernelGGL(wgrad_reduce_kernel,
                      dim3((int)std::min((MN + 255) / 256, (long)4096)),
                      dim3(256), 0, stream, part.data_ptr<float>(),
                      reinterpret_cast<__bf16*>(dw.data_ptr()), MN, d.N,
-                     (int)(R * S * C), splits, Cpad, C);
+                     (int)(R * S * C), atomic ? 1 : splits, Cpad, C);
   return dw;
 }

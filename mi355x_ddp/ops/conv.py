@@ -162,8 +162,9 @@ class _ConvIGEMM(torch.autograd.Function):
             # two-phase: pick the tile (v1/v2 variants vs MIOpen), then the
             # split-K factor for the winning native tile
             kt = ("wgrad_tile", dy.shape, x.shape, R, ctx.stride, ctx.padding)
-            tile = _tuned_choice(kt, ((0, 0), (1, 0), (2, 0), (3, 0), MIOPEN),
-                                 run_dw, default=(0, 0))
+            tile = _tuned_choice(
+                kt, ((0, 0), (1, 0), (2, 0), (3, 0), (4, 0), MIOPEN),
+                run_dw, default=(0, 0))
             if tile == MIOPEN:
                 choice = MIOPEN
             else:
