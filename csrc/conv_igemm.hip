@@ -100,20 +100,26 @@ __device__ __forceinline__ void decode_m2(int m, const ConvDims& d,
 // PADC: gathered tensor has GC==4 (channel-padded stem); d.K is R*S*4 rounded
 // up to BK and the B operand is the [N][d.K] zero-padded weight, so every
 // A-tile tap is one aligned dwordx2 and every B row a pair of float4s.
-template <int MODE, bool FAST, int BMT = BM, bool PADC = false>
+// BNT = GEMM-N tile: 64 default; 128 halves the number of N-tiles — and
+// with it the per-pass re-reads of the gathered A operand — for the wide-N
+// high-C shapes (ResNet50's 1x1 expansions), where fwd/dgrad is A-traffic
+// bound.
+template <int MODE, bool FAST, int BMT = BM, bool PADC = false, int BNT = BN>
 __global__ __launch_bounds__(256)
 void conv_igemm_kernel(const __bf16* __restrict__ Ag,
                        const __bf16* __restrict__ Bg,
                        __bf16* __restrict__ out, ConvDims d) {
   constexpr int MI = BMT / 32;
+  constexpr int NI = BNT / 32;
   constexpr int EPT = BMT / 4;          // 32 or 16 elems per loader thread
   constexpr int PER_ROW = 64 / EPT;     // loader threads per A row
+  constexpr int BEPT = BK * BNT / 256;  // B elems per loader thread
   __shared__ __bf16 sA[BMT * LDK];
-  __shared__ __bf16 sB[BN * LDK];
+  __shared__ __bf16 sB[BNT * LDK];
 
   const int tid = threadIdx.x;
   const int m0 = blockIdx.y * BMT;
-  const int n0 = blockIdx.x * BN;
+  const int n0 = blockIdx.x * BNT;
   const int pa = (MODE == 2) ? (int)(blockIdx.z & 1) : 0;
   const int pb = (MODE == 2) ? (int)(blockIdx.z >> 1) : 0;
 
@@ -127,19 +133,19 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
     if (MODE == 2) decode_m2(m, d, pa, pb, a_n, a_oh, a_ow);
     else decode_m(m, d, a_n, a_oh, a_ow);
   }
-  // B loader: thread -> (row, 16-element quarter of the BK chunk)
-  const int b_row = tid & 63;
-  const int b_off = (tid >> 6) * 16;
+  // B loader: thread -> (row, BEPT-element slice of the BK chunk)
+  const int b_row = tid & (BNT - 1);
+  const int b_off = (tid / BNT) * BEPT;
   const long b_base = (long)min(n0 + b_row, d.N - 1) * d.K;
 
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int wm = (wave >> 1) * (BMT / 2);
-  const int wn = (wave & 1) * 32;
+  const int wn = (wave & 1) * (BNT / 2);
   const int fr = lane & 15;
   const int fk = (lane >> 4) * 8;
 
-  f32x4 acc[MI][2] = {};
+  f32x4 acc[MI][NI] = {};
 
   const int SC = d.S * d.GC;
 
@@ -174,30 +180,30 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
       dst[3] = v3;
     }
   };
-  auto issue_B = [&](int kk0, float4& v0, float4& v1) {
+  auto issue_B = [&](int kk0, float4 (&v)[BEPT / 8]) {
     const float4* src = (const float4*)(Bg + b_base + kk0 + b_off);
-    v0 = src[0];
-    v1 = src[1];
+    #pragma unroll
+    for (int i = 0; i < BEPT / 8; ++i) v[i] = src[i];
   };
-  auto write_B = [&](float4 v0, float4 v1) {
+  auto write_B = [&](const float4 (&v)[BEPT / 8]) {
     float4* dst = (float4*)(sB + b_row * LDK + b_off);
-    dst[0] = v0;
-    dst[1] = v1;
+    #pragma unroll
+    for (int i = 0; i < BEPT / 8; ++i) dst[i] = v[i];
   };
   auto mfma_tile = [&]() {
     #pragma unroll
     for (int ks = 0; ks < BK; ks += 32) {
-      bf16x8 af[MI], bf[2];
+      bf16x8 af[MI], bf[NI];
       #pragma unroll
       for (int mi = 0; mi < MI; ++mi)
         af[mi] = *(const bf16x8*)&sA[(wm + mi * 16 + fr) * LDK + ks + fk];
       #pragma unroll
-      for (int ni = 0; ni < 2; ++ni)
+      for (int ni = 0; ni < NI; ++ni)
         bf[ni] = *(const bf16x8*)&sB[(wn + ni * 16 + fr) * LDK + ks + fk];
       #pragma unroll
       for (int mi = 0; mi < MI; ++mi)
         #pragma unroll
-        for (int ni = 0; ni < 2; ++ni)
+        for (int ni = 0; ni < NI; ++ni)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
     }
@@ -224,9 +230,9 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
       }
       // B tile: rows are d.K-long (BK multiple) -> always-aligned float4s
       {
-        float4 b0, b1;
-        issue_B(kk0, b0, b1);
-        write_B(b0, b1);
+        float4 bv[BEPT / 8];
+        issue_B(kk0, bv);
+        write_B(bv);
       }
       __syncthreads();
       mfma_tile();
@@ -237,25 +243,25 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
     // MFMAs, so HBM/L2 latency overlaps compute; the waits land at the LDS
     // write after the barrier (write-after-barrier form).
     {
-      float4 a0, a1, a2, a3, b0, b1;
+      float4 a0, a1, a2, a3, bv[BEPT / 8];
       issue_A(0, a0, a1, a2, a3);
-      issue_B(0, b0, b1);
+      issue_B(0, bv);
       write_A(a0, a1, a2, a3);
-      write_B(b0, b1);
+      write_B(bv);
     }
     __syncthreads();
     for (int kk0 = 0; kk0 < d.K; kk0 += BK) {
       const bool has_next = kk0 + BK < d.K;
-      float4 a0 = {}, a1 = {}, a2 = {}, a3 = {}, b0 = {}, b1 = {};
+      float4 a0 = {}, a1 = {}, a2 = {}, a3 = {}, bv[BEPT / 8] = {};
       if (has_next) {
         issue_A(kk0 + BK, a0, a1, a2, a3);
-        issue_B(kk0 + BK, b0, b1);
+        issue_B(kk0 + BK, bv);
       }
       mfma_tile();
       if (has_next) {
         __syncthreads();
         write_A(a0, a1, a2, a3);
-        write_B(b0, b1);
+        write_B(bv);
         __syncthreads();
       }
     }
@@ -273,11 +279,11 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
           const int tb = pb + d.pad - d.S + 1 + j;
           if ((ta & 1) || (tb & 1)) continue;
         }
-        float4 a0, a1, a2, a3, b0, b1;
+        float4 a0, a1, a2, a3, bv[BEPT / 8];
         issue_A(kk0, a0, a1, a2, a3);
-        issue_B(kk0, b0, b1);
+        issue_B(kk0, bv);
         write_A(a0, a1, a2, a3);
-        write_B(b0, b1);
+        write_B(bv);
       } else {
         // generic gather: one element at a time (stem conv only)
         for (int e = tid; e < BMT * BK; e += 256) {
@@ -297,7 +303,7 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
           }
           sA[row * LDK + (e & 63)] = v;
         }
-        for (int e = tid; e < BN * BK; e += 256) {
+        for (int e = tid; e < BNT * BK; e += 256) {
           const int row = e >> 6, kk = kk0 + (e & 63);
           sB[row * LDK + (e & 63)] = (kk < d.K)
               ? Bg[(long)min(n0 + row, d.N - 1) * d.K + kk] : (__bf16)0.f;
@@ -315,7 +321,7 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
   #pragma unroll
   for (int mi = 0; mi < MI; ++mi) {
     #pragma unroll
-    for (int ni = 0; ni < 2; ++ni) {
+    for (int ni = 0; ni < NI; ++ni) {
       const int n = n0 + wn + ni * 16 + dn;
       if (n >= d.N) continue;
       #pragma unroll
@@ -865,6 +871,14 @@ at::Tensor conv_fwd_igemm(at::Tensor x, at::Tensor w, long stride, long pad,
   ConvDims d{Nb, P, Q, H, W, C, R, S, (int)stride, (int)pad,
              Nb * P * Q, K, R * S * C};
   const bool fast = (C % BK == 0);
+  if (tile == 228 && fast && d.N >= 96) {
+    // 128x128 tile: halves the N-tile count and with it the A re-reads
+    const dim3 grid((d.N + 127) / 128, (d.M + 127) / 128);
+    hipLaunchKernelGGL((conv_igemm_kernel<0, true, 128, false, 128>), grid,
+                       dim3(256), 0, stream, bf16_ptr(x), bf16_ptr(w),
+                       reinterpret_cast<__bf16*>(out.data_ptr()), d);
+    return out;
+  }
   const int bmt = pick_bmt(d, tile);
   const dim3 grid((d.N + BN - 1) / BN, (d.M + bmt - 1) / bmt);
   auto* kern = bmt == 64
@@ -915,6 +929,13 @@ at::Tensor conv_dgrad_igemm(at::Tensor dy, at::Tensor wT, long H, long W,
     hipLaunchKernelGGL(kern, grid, dim3(256), 0,
                        conv_stream(), bf16_ptr(dy), bf16_ptr(wT),
                        reinterpret_cast<__bf16*>(dx.data_ptr()), d2);
+    return dx;
+  }
+  if (tile == 228 && fast && d.N >= 96) {
+    const dim3 grid((d.N + 127) / 128, (d.M + 127) / 128);
+    hipLaunchKernelGGL((conv_igemm_kernel<1, true, 128, false, 128>), grid,
+                       dim3(256), 0, conv_stream(), bf16_ptr(dy), bf16_ptr(wT),
+                       reinterpret_cast<__bf16*>(dx.data_ptr()), d);
     return dx;
   }
   const int bmt = pick_bmt(d, tile);

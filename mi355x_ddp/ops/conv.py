@@ -119,7 +119,7 @@ class _ConvIGEMM(torch.autograd.Function):
             return C.conv_fwd_igemm(x, weight, stride, padding, t)
 
         key = ("fwd", x.shape, weight.shape, stride, padding)
-        return run(_tuned_choice(key, (64, 128, MIOPEN), run, default=0))
+        return run(_tuned_choice(key, (64, 128, 228, MIOPEN), run, default=0))
 
     @staticmethod
     def backward(ctx, dy):
@@ -141,7 +141,7 @@ class _ConvIGEMM(torch.autograd.Function):
                                           ctx.stride, ctx.padding, t)
 
             key = ("dgrad", dy.shape, wT.shape, ctx.stride, ctx.padding)
-            dx = run_dx(_tuned_choice(key, (64, 128, MIOPEN), run_dx,
+            dx = run_dx(_tuned_choice(key, (64, 128, 228, MIOPEN), run_dx,
                                       default=0))
         if ctx.needs_input_grad[1]:
             R = weight.shape[2]

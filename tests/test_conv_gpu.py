@@ -130,6 +130,26 @@ def test_conv_wgrad_tile_variants_agree(wtile):
             f"{shape} wtile={wtile}: {_rel_err(got, ref.float()):.4f}"
 
 
+def test_conv_fwd_dgrad_tile228_agree():
+    """128x128 tile (BNT=128) fwd/dgrad match the 64-wide-N tile."""
+    from mi355x_ddp import _C
+    for shape in [(8, 64, 32, 32, 256, 1, 1, 0),
+                  (8, 128, 16, 16, 128, 3, 1, 1),
+                  (3, 256, 8, 8, 512, 3, 2, 1)]:
+        x, w, stride, pad = _mk(shape)
+        xb, wb = x.bfloat16(), w.bfloat16()
+        a = _C.conv_fwd_igemm(xb, wb, stride, pad, 128)
+        b = _C.conv_fwd_igemm(xb, wb, stride, pad, 228)
+        assert torch.equal(a, b), shape
+        y = F.conv2d(x, w, None, stride, pad)
+        dy = torch.randn_like(y).to(memory_format=torch.channels_last).bfloat16()
+        wT = wb.contiguous(memory_format=torch.channels_last) \
+            .flip(2, 3).permute(1, 2, 3, 0).contiguous()
+        da = _C.conv_dgrad_igemm(dy, wT, x.shape[2], x.shape[3], stride, pad, 128)
+        db = _C.conv_dgrad_igemm(dy, wT, x.shape[2], x.shape[3], stride, pad, 228)
+        assert torch.equal(da, db), shape
+
+
 def test_stem_fwd_tile_variants_agree():
     """Stem PADC path: both GEMM-M tiles produce the same output."""
     from mi355x_ddp import _C

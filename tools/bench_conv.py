@@ -127,7 +127,7 @@ def main():
         # fwd
         tm = timeit(lambda: F.conv2d(x, wt, None, stride, pad), **kw)
         ti = timeit(lambda: _C.conv_fwd_igemm(x, wt, stride, pad), **kw)
-        tb, cfg = sweep((64, 128),
+        tb, cfg = sweep((64, 128, 228),
                         lambda t: _C.conv_fwd_igemm(x, wt, stride, pad, t))
         row("fwd", tm, ti, tb, cfg)
         # dgrad (skip stem: input grad never needed there)
@@ -138,7 +138,7 @@ def main():
                 [0, 0], 1, [True, False, False]), **kw)
             ti = timeit(lambda: _C.conv_dgrad_igemm(dy, wT, h, w, stride, pad),
                         **kw)
-            tb, cfg = sweep((64, 128), lambda t: _C.conv_dgrad_igemm(
+            tb, cfg = sweep((64, 128, 228), lambda t: _C.conv_dgrad_igemm(
                 dy, wT, h, w, stride, pad, t))
             row("dgrad", tm, ti, tb, cfg)
         # wgrad: sweep tile variants x split factors
