@@ -551,11 +551,14 @@ void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
   constexpr int TN = WGN * 64;          // rsc tile
   __shared__ __bf16 sA[TM * LDK];       // [kout][npq]
   __shared__ __bf16 sB[TN * LDK];       // [rsc][npq]
-  // per-chunk row metadata: for npq row r, the x base offset of tap (0,0)
-  // and the (ih0, iw0) coords for bounds tests (computed once per chunk by
-  // threads 0..63 instead of 4x per staged 4x4 block)
-  __shared__ int sRowOff[64];           // ((xn*GH + ih0)*GW + iw0) or -1
+  // per-chunk row metadata: for npq row r, the x base offset of tap (0,0),
+  // the (ih0, iw0) coords for bounds tests, and a validity flag (computed
+  // once per chunk by threads 0..63 instead of 4x per staged 4x4 block).
+  // NOTE the flag is separate: the tap-(0,0) offset is legitimately
+  // negative for boundary rows of image 0 (ih0/iw0 = -pad).
+  __shared__ int sRowOff[64];           // (xn*GH + ih0)*GW + iw0
   __shared__ short sIh0[64], sIw0[64];
+  __shared__ unsigned char sOk[64];     // npq < d.K
 
   const int tid = threadIdx.x;
   const int m0 = blockIdx.y * TM;
@@ -577,8 +580,9 @@ void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
     // ---- row metadata (threads 0..63) --------------------------------
     if (tid < 64) {
       const int npq = kk0 + tid;
-      int off = -1, ih0 = 0, iw0 = 0;
-      if (npq < d.K) {
+      int off = 0, ih0 = 0, iw0 = 0;
+      const bool ok = npq < d.K;
+      if (ok) {
         int xn, xp, xq;
         decode_m(npq, d, xn, xp, xq);
         ih0 = xp * d.stride - d.pad;
@@ -588,17 +592,26 @@ void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
       sRowOff[tid] = off;
       sIh0[tid] = (short)ih0;
       sIw0[tid] = (short)iw0;
+      sOk[tid] = ok;
     }
     __syncthreads();
 
     // ---- dy tile: TM x 64, 4x4-register-transposed -------------------
+    // thread -> 4x4 block mapping is COLUMN-group-fastest so a wave's
+    // global loads are contiguous (adjacent lanes read adjacent 8 B of the
+    // same dy row). The transposed b64 stores then land on clashing banks
+    // (4-row stride x 36 dwords = same bank pair), so the npq quad is
+    // XOR-swizzled by the row's 4x4-block index — even-only bits keep
+    // 8-element groups intact for the b128 fragment reads, which apply the
+    // same swizzle.
     {
-      constexpr int TA = TM / 64;       // 4x4 blocks per thread
+      constexpr int C4N = TM / 4;       // 4-col groups per row quad
+      constexpr int TA = TM / 64;       // staging passes
       #pragma unroll
       for (int t = 0; t < TA; ++t) {
-        const int bid = tid * TA + t;
-        const int c4 = bid >> 4;        // kout 4-col group
-        const int r4 = bid & 15;        // npq 4-row group
+        const int bid = t * 256 + tid;
+        const int c4 = bid & (C4N - 1); // kout 4-col group
+        const int r4 = bid / C4N;       // npq 4-row group (0..15)
         const int npq0 = kk0 + r4 * 4;
         const int mcol = m0 + c4 * 4;
         uint2 in[4], out[4];
@@ -620,18 +633,22 @@ void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
         }
         tr4x4_bf16(in, out);
         #pragma unroll
-        for (int i = 0; i < 4; ++i)
-          *(uint2*)(sA + (c4 * 4 + i) * LDK + r4 * 4) = out[i];
+        for (int i = 0; i < 4; ++i) {
+          const int row = c4 * 4 + i;
+          const int q = r4 ^ (c4 & 14);
+          *(uint2*)(sA + row * LDK + q * 4) = out[i];
+        }
       }
     }
     // ---- x tile: TN x 64, gather via row metadata --------------------
     {
+      constexpr int C4N = TN / 4;
       constexpr int TB = TN / 64;
       #pragma unroll
       for (int t = 0; t < TB; ++t) {
-        const int bid = tid * TB + t;
-        const int c4 = bid >> 4;
-        const int r4 = bid & 15;
+        const int bid = t * 256 + tid;
+        const int c4 = bid & (C4N - 1);
+        const int r4 = bid / C4N;
         const int nn = n0 + c4 * 4;     // rsc col of this 4-col group
         // FAST contract (C % 64 == 0): nn..nn+3 sit inside ONE filter tap
         const int i_tap = nn / SC;
@@ -644,7 +661,7 @@ void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
         for (int i = 0; i < 4; ++i) {
           const int r = r4 * 4 + i;
           uint2 v = {0u, 0u};
-          if (ncol_ok && sRowOff[r] >= 0) {
+          if (ncol_ok && sOk[r]) {
             const int ih = sIh0[r] + i_tap;
             const int iw = sIw0[r] + j_tap;
             if (ih >= 0 && ih < d.GH && iw >= 0 && iw < d.GW)
@@ -655,8 +672,11 @@ void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
         }
         tr4x4_bf16(in, out);
         #pragma unroll
-        for (int i = 0; i < 4; ++i)
-          *(uint2*)(sB + (c4 * 4 + i) * LDK + r4 * 4) = out[i];
+        for (int i = 0; i < 4; ++i) {
+          const int row = c4 * 4 + i;
+          const int q = r4 ^ (c4 & 14);
+          *(uint2*)(sB + row * LDK + q * 4) = out[i];
+        }
       }
     }
     __syncthreads();
@@ -665,11 +685,17 @@ void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
     for (int ks = 0; ks < BK; ks += 32) {
       bf16x8 af[4], bf[4];
       #pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
-        af[mi] = *(const bf16x8*)&sA[(wm + mi * 16 + fr) * LDK + ks + fk];
+      for (int mi = 0; mi < 4; ++mi) {
+        const int row = wm + mi * 16 + fr;
+        const int q = ((ks + fk) >> 2) ^ ((row >> 2) & 14);
+        af[mi] = *(const bf16x8*)&sA[row * LDK + q * 4];
+      }
       #pragma unroll
-      for (int ni = 0; ni < 4; ++ni)
-        bf[ni] = *(const bf16x8*)&sB[(wn + ni * 16 + fr) * LDK + ks + fk];
+      for (int ni = 0; ni < 4; ++ni) {
+        const int row = wn + ni * 16 + fr;
+        const int q = ((ks + fk) >> 2) ^ ((row >> 2) & 14);
+        bf[ni] = *(const bf16x8*)&sB[row * LDK + q * 4];
+      }
       #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
         #pragma unroll
