@@ -720,10 +720,19 @@ __global__ void gap_fwd_nhwc_kernel(const scalar_t* __restrict__ x,
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   const int n = blockIdx.y;
   if (c >= C) return;
-  float s = 0.f;
+  // 4 partial sums -> 4 loads in flight (the add chain otherwise
+  // serialises one HBM-latency per pixel)
+  float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
   const scalar_t* base = x + (long)n * S * C + c;
-  for (int p = 0; p < S; ++p) s += (float)base[(long)p * C];
-  y[(long)n * C + c] = (scalar_t)(s / S);
+  int p = 0;
+  for (; p + 4 <= S; p += 4) {
+    s0 += (float)base[(long)p * C];
+    s1 += (float)base[(long)(p + 1) * C];
+    s2 += (float)base[(long)(p + 2) * C];
+    s3 += (float)base[(long)(p + 3) * C];
+  }
+  for (; p < S; ++p) s0 += (float)base[(long)p * C];
+  y[(long)n * C + c] = (scalar_t)((s0 + s1 + s2 + s3) / S);
 }
 
 template <typename scalar_t>
