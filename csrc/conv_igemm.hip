@@ -180,15 +180,26 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
       dst[3] = v3;
     }
   };
-  auto issue_B = [&](int kk0, float4 (&v)[BEPT / 8]) {
+  // explicit scalar refs, NOT an array or struct: anything indexable (or
+  // aggregate) here is demoted to scratch — the r2c3 BNT refactor's
+  // float4[] cost the pipelined kernels ~1.6x (144 B/lane spill)
+  auto issue_B = [&](int kk0, float4& v0, float4& v1, float4& v2, float4& v3) {
     const float4* src = (const float4*)(Bg + b_base + kk0 + b_off);
-    #pragma unroll
-    for (int i = 0; i < BEPT / 8; ++i) v[i] = src[i];
+    v0 = src[0];
+    v1 = src[1];
+    if constexpr (BEPT == 32) {
+      v2 = src[2];
+      v3 = src[3];
+    }
   };
-  auto write_B = [&](const float4 (&v)[BEPT / 8]) {
+  auto write_B = [&](float4 v0, float4 v1, float4 v2, float4 v3) {
     float4* dst = (float4*)(sB + b_row * LDK + b_off);
-    #pragma unroll
-    for (int i = 0; i < BEPT / 8; ++i) dst[i] = v[i];
+    dst[0] = v0;
+    dst[1] = v1;
+    if constexpr (BEPT == 32) {
+      dst[2] = v2;
+      dst[3] = v3;
+    }
   };
   auto mfma_tile = [&]() {
     #pragma unroll
@@ -230,9 +241,9 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
       }
       // B tile: rows are d.K-long (BK multiple) -> always-aligned float4s
       {
-        float4 bv[BEPT / 8];
-        issue_B(kk0, bv);
-        write_B(bv);
+        float4 b0, b1, b2, b3;
+        issue_B(kk0, b0, b1, b2, b3);
+        write_B(b0, b1, b2, b3);
       }
       __syncthreads();
       mfma_tile();
@@ -243,25 +254,26 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
     // MFMAs, so HBM/L2 latency overlaps compute; the waits land at the LDS
     // write after the barrier (write-after-barrier form).
     {
-      float4 a0, a1, a2, a3, bv[BEPT / 8];
+      float4 a0, a1, a2, a3, b0, b1, b2, b3;
       issue_A(0, a0, a1, a2, a3);
-      issue_B(0, bv);
+      issue_B(0, b0, b1, b2, b3);
       write_A(a0, a1, a2, a3);
-      write_B(bv);
+      write_B(b0, b1, b2, b3);
     }
     __syncthreads();
     for (int kk0 = 0; kk0 < d.K; kk0 += BK) {
       const bool has_next = kk0 + BK < d.K;
-      float4 a0 = {}, a1 = {}, a2 = {}, a3 = {}, bv[BEPT / 8] = {};
+      float4 a0 = {}, a1 = {}, a2 = {}, a3 = {};
+      float4 b0 = {}, b1 = {}, b2 = {}, b3 = {};
       if (has_next) {
         issue_A(kk0 + BK, a0, a1, a2, a3);
-        issue_B(kk0 + BK, bv);
+        issue_B(kk0 + BK, b0, b1, b2, b3);
       }
       mfma_tile();
       if (has_next) {
         __syncthreads();
         write_A(a0, a1, a2, a3);
-        write_B(bv);
+        write_B(b0, b1, b2, b3);
         __syncthreads();
       }
     }
@@ -279,11 +291,11 @@ void conv_igemm_kernel(const __bf16* __restrict__ Ag,
           const int tb = pb + d.pad - d.S + 1 + j;
           if ((ta & 1) || (tb & 1)) continue;
         }
-        float4 a0, a1, a2, a3, bv[BEPT / 8];
+        float4 a0, a1, a2, a3, b0, b1, b2, b3;
         issue_A(kk0, a0, a1, a2, a3);
-        issue_B(kk0, bv);
+        issue_B(kk0, b0, b1, b2, b3);
         write_A(a0, a1, a2, a3);
-        write_B(bv);
+        write_B(b0, b1, b2, b3);
       } else {
         // generic gather: one element at a time (stem conv only)
         for (int e = tid; e < BMT * BK; e += 256) {
