@@ -737,6 +737,168 @@ void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
     }
 }
 
+// ---------------------------------------------------------------------------
+// wgrad v3 (LIN): 1x1 stride-1 pad-0 shapes — the bulk of ResNet50's wgrad
+// time. im2col(x) IS x, so BOTH operands are plain row-major [npq][C]:
+// no gather, no row metadata, and the whole chunk pipeline can be
+// register-double-buffered like the fwd kernel (loads for chunk k+1 issue
+// before chunk k's MFMAs, hiding the ~900-cycle HBM latency that left v2 at
+// ~38% of peak bandwidth). Tile = (WGM*64) x (WGN*64), 2 barriers/chunk.
+// ---------------------------------------------------------------------------
+template <int WGM, int WGN, bool DIRECT>
+__global__ __launch_bounds__(256, 2)  // 2 waves/SIMD: the double buffer
+                                      // needs a partner wave per SIMD
+void conv_wgrad_lin_kernel(const __bf16* __restrict__ dy,
+                           const __bf16* __restrict__ x,
+                           float* __restrict__ dwp,
+                           __bf16* __restrict__ dwb, ConvDims d) {
+  constexpr int TM = WGM * 64;
+  constexpr int TN = WGN * 64;
+  constexpr int TA = TM / 64;
+  constexpr int TB = TN / 64;
+  __shared__ __bf16 sA[TM * LDK];
+  __shared__ __bf16 sB[TN * LDK];
+
+  const int tid = threadIdx.x;
+  const int m0 = blockIdx.y * TM;
+  const int n0 = blockIdx.x * TN;
+
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wm = (wave / WGN) * 64;
+  const int wn = (wave % WGN) * 64;
+  const int fr = lane & 15;
+  const int fk = (lane >> 4) * 8;
+
+  const int nchunks = (d.K + BK - 1) / BK;
+  f32x4 acc[4][4] = {};
+
+  // staged 4x4 blocks for one chunk: [block][row] uint2, fully unrolled
+  // (verify ScratchSize stays 0 — indexable locals here have spilled before)
+  uint2 sta[TA][4], stb[TB][4];
+
+  // per-pass block decomposition (c4-fastest => coalesced loads)
+  auto load_op = [&](const __bf16* src, int ncols, int c4n, int base,
+                     int kk0, int bid, uint2 (&in)[4]) {
+    const int c4 = bid & (c4n - 1);
+    const int r4 = bid / c4n;
+    const int row0 = kk0 + r4 * 4;
+    const int col = base + c4 * 4;
+    if (row0 + 4 <= d.K && col + 4 <= ncols) {
+      #pragma unroll
+      for (int i = 0; i < 4; ++i)
+        in[i] = *(const uint2*)(src + (long)(row0 + i) * ncols + col);
+    } else {
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        __bf16 e[4] = {};
+        if (row0 + i < d.K)
+          #pragma unroll
+          for (int c = 0; c < 4; ++c)
+            if (col + c < ncols) e[c] = src[(long)(row0 + i) * ncols + col + c];
+        in[i] = *(uint2*)e;
+      }
+    }
+  };
+  auto issue_all = [&](int kk0) {
+    #pragma unroll
+    for (int t = 0; t < TA; ++t)
+      load_op(dy, d.M, TM / 4, m0, kk0, t * 256 + tid, sta[t]);
+    #pragma unroll
+    for (int t = 0; t < TB; ++t)
+      load_op(x, d.N, TN / 4, n0, kk0, t * 256 + tid, stb[t]);
+  };
+  auto write_all = [&]() {
+    #pragma unroll
+    for (int t = 0; t < TA; ++t) {
+      const int bid = t * 256 + tid;
+      const int c4 = bid & (TM / 4 - 1);
+      const int r4 = bid / (TM / 4);
+      uint2 out[4];
+      tr4x4_bf16(sta[t], out);
+      const int q = r4 ^ (c4 & 14);
+      #pragma unroll
+      for (int i = 0; i < 4; ++i)
+        *(uint2*)(sA + (c4 * 4 + i) * LDK + q * 4) = out[i];
+    }
+    #pragma unroll
+    for (int t = 0; t < TB; ++t) {
+      const int bid = t * 256 + tid;
+      const int c4 = bid & (TN / 4 - 1);
+      const int r4 = bid / (TN / 4);
+      uint2 out[4];
+      tr4x4_bf16(stb[t], out);
+      const int q = r4 ^ (c4 & 14);
+      #pragma unroll
+      for (int i = 0; i < 4; ++i)
+        *(uint2*)(sB + (c4 * 4 + i) * LDK + q * 4) = out[i];
+    }
+  };
+  auto mfma_all = [&]() {
+    #pragma unroll
+    for (int ks = 0; ks < BK; ks += 32) {
+      bf16x8 af[4], bf[4];
+      #pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        const int row = wm + mi * 16 + fr;
+        const int q = ((ks + fk) >> 2) ^ ((row >> 2) & 14);
+        af[mi] = *(const bf16x8*)&sA[row * LDK + q * 4];
+      }
+      #pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const int row = wn + ni * 16 + fr;
+        const int q = ((ks + fk) >> 2) ^ ((row >> 2) & 14);
+        bf[ni] = *(const bf16x8*)&sB[row * LDK + q * 4];
+      }
+      #pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    }
+  };
+
+  bool first = true;
+  for (int kc = blockIdx.z; kc < nchunks; kc += gridDim.z) {
+    const int kk0 = kc * BK;
+    if (first) {
+      issue_all(kk0);
+      write_all();
+      __syncthreads();
+      first = false;
+    }
+    const int kn = kc + gridDim.z;
+    const bool has_next = kn < nchunks;
+    // chunk kn's global loads issue before chunk kc's MFMAs (latency cover)
+    if (has_next) issue_all(kn * BK);
+    mfma_all();
+    if (has_next) {
+      __syncthreads();
+      write_all();
+      __syncthreads();
+    }
+  }
+
+  const int dm = (lane >> 4) * 4;
+  const int dn = lane & 15;
+  float* slab = DIRECT ? nullptr : dwp + (long)blockIdx.z * d.M * d.N;
+  #pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+    #pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      const int n = n0 + wn + ni * 16 + dn;
+      if (n >= d.N) continue;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = m0 + wm + mi * 16 + dm + r;
+        if (m >= d.M) continue;
+        if (DIRECT) dwb[(long)m * d.N + n] = (__bf16)acc[mi][ni][r];
+        else slab[(long)m * d.N + n] = acc[mi][ni][r];
+      }
+    }
+}
+
 // Sum the split-K partial slabs and emit the bf16 channels_last weight grad:
 // out[k][tap*Cout + c] (= memory layout of channels_last (K, Cout, R, S)).
 // Cpad==Cout -> identity column map; the stem maps Cpad=4 -> Cout=3 and drops
@@ -1024,16 +1186,25 @@ at::Tensor conv_wgrad_igemm(at::Tensor dy, at::Tensor x, long R, long S,
 
   // tile pick: v2's larger tiles read dy/x fewer times (traffic per pass
   // scales with the tile count along the other axis) and stage with wide
-  // LDS writes; v1 remains for small-M shapes and the generic/stem paths
+  // LDS writes; LIN (1x1 s1 p0: both operands plain row-major) adds the
+  // register-pipelined double buffer; v1 remains for small-M shapes and
+  // the generic/stem paths
+  const bool lin_ok = fast && !stem && R == 1 && S == 1 &&
+                      stride == 1 && pad == 0;
   int tile = (int)wtile;
   if (tile == 0) {
-    if (fast && !stem && d.M >= 256 && d.N <= 64) tile = 3;
+    if (lin_ok && d.M >= 128 && d.N >= 96) tile = 5;
+    else if (lin_ok && d.M >= 256 && d.N <= 64) tile = 6;
+    else if (fast && !stem && d.M >= 256 && d.N <= 64) tile = 3;
     else if (fast && !stem && d.M >= 128 && d.N >= 96) tile = 2;
     else tile = 1;
   }
+  if ((tile == 5 || tile == 6) && !lin_ok) tile = tile - 3;  // 2 / 3
   if (!fast || stem) tile = (tile == 4) ? 4 : 1;  // v2 needs the FAST layout
-  const int TM = tile == 3 ? 256 : (tile == 2 ? 128 : 64);
-  const int TN = tile == 2 ? 128 : (tile == 3 ? 64 : 128);
+  const int TM = (tile == 3 || tile == 6) ? 256
+               : (tile == 2 || tile == 5) ? 128 : 64;
+  const int TN = (tile == 2 || tile == 5) ? 128
+               : (tile == 3 || tile == 6) ? 64 : 128;
 
   const int tm = (d.M + TM - 1) / TM, tn = (d.N + TN - 1) / TN;
   const int nchunks = (d.K + BK - 1) / BK;
@@ -1050,7 +1221,9 @@ at::Tensor conv_wgrad_igemm(at::Tensor dy, at::Tensor x, long R, long S,
 
   if (splits == 1 && !stem && tile != 4) {
     auto* kern =
-        tile == 3 ? conv_wgrad_v2_kernel<4, 1, true>
+        tile == 6 ? conv_wgrad_lin_kernel<4, 1, true>
+        : tile == 5 ? conv_wgrad_lin_kernel<2, 2, true>
+        : tile == 3 ? conv_wgrad_v2_kernel<4, 1, true>
         : tile == 2 ? conv_wgrad_v2_kernel<2, 2, true>
         : (fast ? conv_wgrad_kernel<true, false, true>
                 : conv_wgrad_kernel<false, false, true>);
@@ -1065,7 +1238,9 @@ at::Tensor conv_wgrad_igemm(at::Tensor dy, at::Tensor x, long R, long S,
       ? at::zeros({MN}, x.options().dtype(at::kFloat))
       : at::empty({(long)splits * MN}, x.options().dtype(at::kFloat));
   auto* kern =
-      tile == 3 ? conv_wgrad_v2_kernel<4, 1, false>
+      tile == 6 ? conv_wgrad_lin_kernel<4, 1, false>
+      : tile == 5 ? conv_wgrad_lin_kernel<2, 2, false>
+      : tile == 3 ? conv_wgrad_v2_kernel<4, 1, false>
       : tile == 2 ? conv_wgrad_v2_kernel<2, 2, false>
       : atomic ? (stem ? conv_wgrad_kernel<false, true, false, true>
                  : fast ? conv_wgrad_kernel<true, false, false, true>

@@ -163,7 +163,8 @@ class _ConvIGEMM(torch.autograd.Function):
             # split-K factor for the winning native tile
             kt = ("wgrad_tile", dy.shape, x.shape, R, ctx.stride, ctx.padding)
             tile = _tuned_choice(
-                kt, ((0, 0), (1, 0), (2, 0), (3, 0), (4, 0), MIOPEN),
+                kt, ((0, 0), (1, 0), (2, 0), (3, 0), (4, 0), (5, 0), (6, 0),
+                     MIOPEN),
                 run_dw, default=(0, 0))
             if tile == MIOPEN:
                 choice = MIOPEN

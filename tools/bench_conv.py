@@ -146,7 +146,8 @@ def main():
             dy, x, wt, None, [stride, stride], [pad, pad], [1, 1], False,
             [0, 0], 1, [False, True, False]), **kw)
         ti = timeit(lambda: _C.conv_wgrad_igemm(dy, x, r, r, stride, pad), **kw)
-        wcands = [(t, s) for t in (1, 2, 3, 4) for s in (0, 1, 32, 128)] \
+        wcands = [(t, s) for t in (1, 2, 3, 4, 5, 6)
+                  for s in (0, 1, 32, 128)] \
             if c != 3 else [(t, s) for t in (1, 4)
                             for s in (0, 32, 128, 512)]
         tb, cfg = sweep(wcands, lambda ts: _C.conv_wgrad_igemm(
