@@ -397,20 +397,33 @@ __global__ void bn_stats_nhwc_v2_kernel(const scalar_t* __restrict__ x,
   const int r = threadIdx.x / tpr;
   const int c0 = c_idx * 8;
   float s[8] = {}, sq[8] = {};
-  for (long p = (long)blockIdx.y * rpb + r; p < P;
-       p += (long)gridDim.y * rpb) {
-    scalar_t v8[8];
-    if constexpr (sizeof(scalar_t) == 2) {
-      *(float4*)v8 = *(const float4*)(x + p * C + c0);
-    } else {
-      #pragma unroll
-      for (int e = 0; e < 8; ++e) v8[e] = x[p * C + c0 + e];
+  // 4 rows per loop pass: the loads are issued together (independent) so
+  // up to 4 HBM fetches overlap instead of one latency per row
+  const long pstride = (long)gridDim.y * rpb;
+  for (long p0 = (long)blockIdx.y * rpb + r; p0 < P; p0 += 4 * pstride) {
+    scalar_t v8[4][8];
+    bool ok[4];
+    #pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      const long p = p0 + k * pstride;
+      ok[k] = p < P;
+      if (!ok[k]) continue;
+      if constexpr (sizeof(scalar_t) == 2) {
+        *(float4*)v8[k] = *(const float4*)(x + p * C + c0);
+      } else {
+        #pragma unroll
+        for (int e = 0; e < 8; ++e) v8[k][e] = x[p * C + c0 + e];
+      }
     }
     #pragma unroll
-    for (int e = 0; e < 8; ++e) {
-      const float v = (float)v8[e];
-      s[e] += v;
-      sq[e] += v * v;
+    for (int k = 0; k < 4; ++k) {
+      if (!ok[k]) continue;
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const float v = (float)v8[k][e];
+        s[e] += v;
+        sq[e] += v * v;
+      }
     }
   }
   #pragma unroll
@@ -498,28 +511,40 @@ __global__ void bn_bwd_reduce_nhwc_v2_kernel(const scalar_t* __restrict__ dy,
     is[e] = invstd[c0 + e];
   }
   float sdy[8] = {}, sdyx[8] = {};
-  for (long p = (long)blockIdx.y * rpb + r; p < P;
-       p += (long)gridDim.y * rpb) {
-    const long base = p * C + c0;
-    scalar_t dy8[8], x8[8], y8[8];
-    if constexpr (sizeof(scalar_t) == 2) {
-      *(float4*)dy8 = *(const float4*)(dy + base);
-      *(float4*)x8 = *(const float4*)(x + base);
-      if (RELU) *(float4*)y8 = *(const float4*)(y + base);
-    } else {
-      #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        dy8[e] = dy[base + e];
-        x8[e] = x[base + e];
-        if (RELU) y8[e] = y[base + e];
+  // 2 rows per pass (2-3 tensors each): 4-6 fetches in flight
+  const long pstride = (long)gridDim.y * rpb;
+  for (long p0 = (long)blockIdx.y * rpb + r; p0 < P; p0 += 2 * pstride) {
+    scalar_t dy8[2][8], x8[2][8], y8[2][8];
+    bool ok[2];
+    #pragma unroll
+    for (int k = 0; k < 2; ++k) {
+      const long p = p0 + k * pstride;
+      ok[k] = p < P;
+      if (!ok[k]) continue;
+      const long base = p * C + c0;
+      if constexpr (sizeof(scalar_t) == 2) {
+        *(float4*)dy8[k] = *(const float4*)(dy + base);
+        *(float4*)x8[k] = *(const float4*)(x + base);
+        if (RELU) *(float4*)y8[k] = *(const float4*)(y + base);
+      } else {
+        #pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          dy8[k][e] = dy[base + e];
+          x8[k][e] = x[base + e];
+          if (RELU) y8[k][e] = y[base + e];
+        }
       }
     }
     #pragma unroll
-    for (int e = 0; e < 8; ++e) {
-      float g = (float)dy8[e];
-      if (RELU && (float)y8[e] <= 0.f) g = 0.f;
-      sdy[e] += g;
-      sdyx[e] += g * ((float)x8[e] - mu[e]) * is[e];
+    for (int k = 0; k < 2; ++k) {
+      if (!ok[k]) continue;
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float g = (float)dy8[k][e];
+        if (RELU && (float)y8[k][e] <= 0.f) g = 0.f;
+        sdy[e] += g;
+        sdyx[e] += g * ((float)x8[k][e] - mu[e]) * is[e];
+      }
     }
   }
   #pragma unroll
