@@ -151,8 +151,15 @@ def main():
 
     # --- warmup ------------------------------------------------------------
     model.train()
+    import sys
     for i in range(args.warmup):
+        t_w = time.perf_counter()
         one_step(i)
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+        if rank == 0:
+            print(f"[warmup {i}] {time.perf_counter() - t_w:.2f}s",
+                  file=sys.stderr, flush=True)
 
     # --- timed region ------------------------------------------------------
     if world > 1:
