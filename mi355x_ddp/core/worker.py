@@ -27,17 +27,26 @@ from .checkpoint import load_checkpoint
 from .engine import fit
 
 
-def init_seeds(seed: int, deterministic: bool = False) -> None:
+def init_seeds(seed: int, deterministic: bool = False,
+               benchmark: bool = False) -> None:
     """Per-rank seeding (reference distributed_mp.py:29-39), without the
     reference's quirk of re-enabling benchmark mode right after requesting
-    determinism (distributed_mp.py:61)."""
+    determinism (distributed_mp.py:61).
+
+    `benchmark` defaults OFF (the reference set cudnn.benchmark=True,
+    distributed.py:48): on ROCm that flag switches MIOpen to exhaustive
+    find, which COMPILES candidate kernels per conv shape — minutes per
+    224px shape on a cold cache, measured hanging the first training step.
+    The per-shape autotune cache in ops/conv.py is this framework's
+    benchmark=True equivalent (it also covers the MIOpen immediate-mode
+    kernel as a candidate)."""
     random.seed(seed)
     np.random.seed(seed)
     torch.manual_seed(seed)
     if torch.cuda.is_available():
         torch.cuda.manual_seed_all(seed)
     torch.backends.cudnn.deterministic = deterministic
-    torch.backends.cudnn.benchmark = not deterministic
+    torch.backends.cudnn.benchmark = benchmark and not deterministic
 
 
 def build_training(cfg: TrainConfig, device: torch.device, world_size: int,
