@@ -186,7 +186,10 @@ def main():
     if rank == 0:
         arch_label = {"resnet18": "ResNet18", "resnet34": "ResNet34",
                       "resnet50": "ResNet50", "resnet101": "ResNet101",
-                      "resnet152": "ResNet152"}[args.arch]
+                      "resnet152": "ResNet152",
+                      "resnet18_imagenet": "ResNet18-ImageNet",
+                      "resnet50_imagenet": "ResNet50-ImageNet"}.get(
+                          args.arch, args.arch)
         data_label = "CIFAR-100" if args.image_size == 32 else             f"synthetic-{args.image_size}px"
         result = {
             "metric": f"images/sec, {arch_label} {data_label} DDP training",

@@ -90,7 +90,9 @@ def add_common_args(p: argparse.ArgumentParser) -> argparse.ArgumentParser:
     p.add_argument("--ip", default="127.0.0.1", type=str)
     p.add_argument("--port", default=23456, type=int)
     p.add_argument("--arch", default="resnet18", type=str,
-                   choices=["resnet18", "resnet34", "resnet50", "resnet101", "resnet152"])
+                   choices=["resnet18", "resnet34", "resnet50", "resnet101",
+                            "resnet152", "resnet18_imagenet",
+                            "resnet50_imagenet"])
     p.add_argument("--amp", default=None, type=str, choices=[None, "fp32", "bf16", "fp16"])
     p.add_argument("--no-sync-bn", action="store_true")
     p.add_argument("--num_workers", default=4, type=int)

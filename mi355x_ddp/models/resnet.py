@@ -86,11 +86,22 @@ class Bottleneck(nn.Module):
 
 
 class ResNet(nn.Module):
-    def __init__(self, block, num_blocks, num_classes: int = 100):
+    def __init__(self, block, num_blocks, num_classes: int = 100,
+                 imagenet_stem: bool = False):
         super().__init__()
         self.in_channels = 64
-        # CIFAR stem: single 3x3, stride 1, no maxpool (reference utils/model.py:66-70)
-        self.conv1 = _conv3x3(3, 64)
+        if imagenet_stem:
+            # standard ImageNet stem: 7x7 s2 + 3x3 s2 maxpool (4x spatial
+            # reduction before stage 1) — beyond the reference zoo, for
+            # 224px inputs; the PADC conv path handles the 7x7 C=3 stem
+            self.conv1 = nn.Conv2d(3, 64, kernel_size=7, stride=2,
+                                   padding=3, bias=False)
+            self.maxpool = nn.MaxPool2d(kernel_size=3, stride=2, padding=1)
+        else:
+            # CIFAR stem: single 3x3, stride 1, no maxpool (reference
+            # utils/model.py:66-70)
+            self.conv1 = _conv3x3(3, 64)
+            self.maxpool = None
         self.bn1 = nn.BatchNorm2d(64)
         self.layer1 = self._make_layer(block, 64, num_blocks[0], 1)
         self.layer2 = self._make_layer(block, 128, num_blocks[1], 2)
@@ -109,6 +120,8 @@ class ResNet(nn.Module):
 
     def forward(self, x):
         out = bn_relu(self.conv1(x), self.bn1)
+        if self.maxpool is not None:
+            out = self.maxpool(out)
         out = self.layer1(out)
         out = self.layer2(out)
         out = self.layer3(out)
@@ -140,8 +153,22 @@ def resnet152(num_classes: int = 100) -> ResNet:
     return ResNet(Bottleneck, [3, 8, 36, 3], num_classes)
 
 
+def resnet50_imagenet(num_classes: int = 100) -> ResNet:
+    """ImageNet-topology ResNet50 (7x7 s2 stem + maxpool, stages at
+    56/28/14/7 px for 224px inputs) — beyond the reference zoo, which is
+    CIFAR-only; the BASELINE 'ImageNet-shape' config can run either this
+    or the CIFAR-stem resnet50 on 224px data."""
+    return ResNet(Bottleneck, [3, 4, 6, 3], num_classes, imagenet_stem=True)
+
+
+def resnet18_imagenet(num_classes: int = 100) -> ResNet:
+    return ResNet(BasicBlock, [2, 2, 2, 2], num_classes, imagenet_stem=True)
+
+
 _FACTORIES = {"resnet18": resnet18, "resnet34": resnet34, "resnet50": resnet50,
-              "resnet101": resnet101, "resnet152": resnet152}
+              "resnet101": resnet101, "resnet152": resnet152,
+              "resnet18_imagenet": resnet18_imagenet,
+              "resnet50_imagenet": resnet50_imagenet}
 
 
 def build_model(arch: str, num_classes: int = 100) -> ResNet:

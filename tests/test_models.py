@@ -56,3 +56,23 @@ def test_deep_zoo_shapes():
         assert n_bottleneck == blocks
         y = m(torch.randn(2, 3, 32, 32))
         assert y.shape == (2, 100)
+
+
+def test_imagenet_stem_variants():
+    """Beyond the reference zoo: ImageNet-topology stems (7x7 s2 + maxpool,
+    4x spatial reduction before stage 1)."""
+    import torch
+    from mi355x_ddp.models import build_model
+    m = build_model("resnet50_imagenet")
+    y = m(torch.randn(2, 3, 224, 224))
+    assert y.shape == (2, 100)
+    # stage-1 input is 56x56 for 224px inputs (7x7 s2 -> 112, maxpool -> 56)
+    feats = {}
+    def grab(mod, inp):
+        feats["hw"] = inp[0].shape[-2:]
+    h = m.layer1.register_forward_pre_hook(grab)
+    m(torch.randn(1, 3, 224, 224))
+    h.remove()
+    assert tuple(feats["hw"]) == (56, 56)
+    m18 = build_model("resnet18_imagenet")
+    assert m18(torch.randn(1, 3, 96, 96)).shape == (1, 100)
