@@ -222,6 +222,11 @@ class MI355SyncBatchNorm(nn.BatchNorm2d):
             out = cls(module.num_features, module.eps, module.momentum,
                       module.affine, module.track_running_stats,
                       process_group=process_group)
+            # preserve the source module's device/dtype: converting an
+            # already-.cuda() model must not leave the new BN params on CPU
+            ref = module.weight if module.affine else module.running_mean
+            if ref is not None:
+                out = out.to(device=ref.device, dtype=ref.dtype)
             if module.affine:
                 with torch.no_grad():
                     out.weight.copy_(module.weight)

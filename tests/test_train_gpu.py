@@ -35,7 +35,9 @@ def test_single_gpu_step(amp):
         (scaler.scale_loss(loss) if scaler else loss).backward()
         model.finalize_backward()
         if scaler:
-            scaler.unscale_([model.flat_grads])
+            grads = [model.flat_grads] if model.flat_grads is not None else \
+                [p.grad for p in model.parameters() if p.grad is not None]
+            scaler.unscale_(grads)
             scaler.step(opt)
         else:
             opt.step()
@@ -107,7 +109,9 @@ def _graph_vs_eager(make_model, data, steps_tol):
         cfg = TrainConfig(batch_size=data[0][0].shape[0], amp="fp32",
                           sync_bn=False, hip_graph=True)
         net = make_model().to(device)
-        model = FlatDDP(net, overlap=False)
+        # hipGraph capture needs static grad memory (world-1 default is the
+        # no-view fast path, which reallocates grads per step)
+        model = FlatDDP(net, overlap=False, static_grads=True)
         opt = FusedSGD(model.parameters(), lr=0.05, momentum=0.9)
         model.train()
         snap = [p.detach().clone() for p in model.parameters()]
@@ -239,7 +243,9 @@ def test_grad_accumulation_gpu_step():
     model.finalize_backward()
     opt.step()
     torch.cuda.synchronize()
-    assert torch.isfinite(model.flat_grads).all()
+    grads = [model.flat_grads] if model.flat_grads is not None else \
+        [p.grad for p in model.parameters() if p.grad is not None]
+    assert all(torch.isfinite(g).all() for g in grads)
 
 
 def test_engine_validate_gpu():
