@@ -787,6 +787,108 @@ __global__ void gap_bwd_kernel(const scalar_t* __restrict__ dy,
 }
 
 // ---------------------------------------------------------------------------
+// MaxPool2d NHWC (the ImageNet stem's 3x3 s2 pool). Forward records the
+// window argmax (uint8, first-max tie-break like ATen); backward GATHERS:
+// each input pixel sums the dy of the <=4 overlapping windows that chose
+// it — no atomics, no scatter. Vectorized 16 B over channels.
+// ---------------------------------------------------------------------------
+template <typename scalar_t>
+__global__ void maxpool_fwd_nhwc_kernel(const scalar_t* __restrict__ x,
+                                        scalar_t* __restrict__ y,
+                                        unsigned char* __restrict__ idx,
+                                        int Nb, int H, int W, int C,
+                                        int P, int Q, int K, int S, int pad) {
+  constexpr int V = 16 / sizeof(scalar_t);
+  const long nvec = (long)Nb * P * Q * (C / V);
+  const int cpr = C / V;
+  for (long o = (long)blockIdx.x * blockDim.x + threadIdx.x; o < nvec;
+       o += (long)gridDim.x * blockDim.x) {
+    const int cv = (int)(o % cpr);
+    long t = o / cpr;
+    const int q = (int)(t % Q); t /= Q;
+    const int p = (int)(t % P);
+    const int n = (int)(t / P);
+    const int c0 = cv * V;
+    float best[V];
+    unsigned char bidx[V];
+    #pragma unroll
+    for (int e = 0; e < V; ++e) { best[e] = -3.4e38f; bidx[e] = 0; }
+    const int ih0 = p * S - pad, iw0 = q * S - pad;
+    for (int kh = 0; kh < K; ++kh) {
+      const int ih = ih0 + kh;
+      if (ih < 0 || ih >= H) continue;
+      for (int kw = 0; kw < K; ++kw) {
+        const int iw = iw0 + kw;
+        if (iw < 0 || iw >= W) continue;
+        scalar_t v[16 / sizeof(scalar_t)];
+        *(float4*)v = *(const float4*)(x +
+            (((long)n * H + ih) * W + iw) * C + c0);
+        const unsigned char wi = (unsigned char)(kh * K + kw);
+        #pragma unroll
+        for (int e = 0; e < V; ++e) {
+          const float f = (float)v[e];
+          if (f > best[e]) { best[e] = f; bidx[e] = wi; }
+        }
+      }
+    }
+    scalar_t out[16 / sizeof(scalar_t)];
+    #pragma unroll
+    for (int e = 0; e < V; ++e) out[e] = (scalar_t)best[e];
+    const long ob = (((long)n * P + p) * Q + q) * C + c0;
+    *(float4*)(y + ob) = *(float4*)out;
+    #pragma unroll
+    for (int e = 0; e < V; ++e) idx[ob + e] = bidx[e];
+  }
+}
+
+template <typename scalar_t>
+__global__ void maxpool_bwd_nhwc_kernel(const scalar_t* __restrict__ dy,
+                                        const unsigned char* __restrict__ idx,
+                                        scalar_t* __restrict__ dx,
+                                        int Nb, int H, int W, int C,
+                                        int P, int Q, int K, int S, int pad) {
+  constexpr int V = 16 / sizeof(scalar_t);
+  const long nvec = (long)Nb * H * W * (C / V);
+  const int cpr = C / V;
+  for (long o = (long)blockIdx.x * blockDim.x + threadIdx.x; o < nvec;
+       o += (long)gridDim.x * blockDim.x) {
+    const int cv = (int)(o % cpr);
+    long t = o / cpr;
+    const int iw = (int)(t % W); t /= W;
+    const int ih = (int)(t % H);
+    const int n = (int)(t / H);
+    const int c0 = cv * V;
+    float acc[V];
+    #pragma unroll
+    for (int e = 0; e < V; ++e) acc[e] = 0.f;
+    // windows (p,q) with p*S - pad <= ih < p*S - pad + K
+    const int p_lo = max(0, (ih + pad - K + S) / S);
+    const int p_hi = min(P - 1, (ih + pad) / S);
+    const int q_lo = max(0, (iw + pad - K + S) / S);
+    const int q_hi = min(Q - 1, (iw + pad) / S);
+    for (int p = p_lo; p <= p_hi; ++p) {
+      const int kh = ih - (p * S - pad);
+      if (kh < 0 || kh >= K) continue;
+      for (int q = q_lo; q <= q_hi; ++q) {
+        const int kw = iw - (q * S - pad);
+        if (kw < 0 || kw >= K) continue;
+        const long ob = (((long)n * P + p) * Q + q) * C + c0;
+        const unsigned char wi = (unsigned char)(kh * K + kw);
+        scalar_t g[16 / sizeof(scalar_t)];
+        *(float4*)g = *(const float4*)(dy + ob);
+        #pragma unroll
+        for (int e = 0; e < V; ++e)
+          if (idx[ob + e] == wi) acc[e] += (float)g[e];
+      }
+    }
+    scalar_t out[16 / sizeof(scalar_t)];
+    #pragma unroll
+    for (int e = 0; e < V; ++e) out[e] = (scalar_t)acc[e];
+    *(float4*)(dx + (((long)n * H + ih) * W + iw) * C + c0) = *(float4*)out;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Per-row rank of the target class:
 //   rank = #{j : logit[j] > logit[t]} + #{j < t : logit[j] == logit[t]}.
 // acc@k = mean(rank < k). Ties break by smaller class index (a stable
@@ -1337,6 +1439,54 @@ at::Tensor gap_bwd(at::Tensor dy, at::Tensor x_like) {
   return dx;
 }
 
+std::vector<at::Tensor> maxpool_fwd(at::Tensor x, long k, long stride,
+                                    long pad) {
+  CHECK_CUDA(x);
+  TORCH_CHECK(x.dim() == 4 &&
+              x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "maxpool_fwd expects NHWC (channels_last)");
+  const int Nb = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  TORCH_CHECK(C % 8 == 0, "maxpool_fwd needs C % 8 == 0");
+  const int P = (H + 2 * (int)pad - (int)k) / (int)stride + 1;
+  const int Q = (W + 2 * (int)pad - (int)k) / (int)stride + 1;
+  auto y = at::empty({Nb, C, P, Q},
+                     x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  auto idx = at::empty({(long)Nb * P * Q * C}, x.options().dtype(at::kByte));
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::Half, at::ScalarType::BFloat16,
+      x.scalar_type(), "maxpool_fwd", [&] {
+    const long nvec = (long)Nb * P * Q * C / (16 / sizeof(scalar_t));
+    const int blocks = (int)std::min((nvec + 255) / 256, (long)4096);
+    hipLaunchKernelGGL(maxpool_fwd_nhwc_kernel<scalar_t>,
+                       dim3(std::max(blocks, 1)), dim3(256), 0, cur_stream(),
+                       x.data_ptr<scalar_t>(), y.data_ptr<scalar_t>(),
+                       idx.data_ptr<unsigned char>(), Nb, H, W, C, P, Q,
+                       (int)k, (int)stride, (int)pad);
+  });
+  return {y, idx};
+}
+
+at::Tensor maxpool_bwd(at::Tensor dy, at::Tensor idx, long H, long W,
+                       long k, long stride, long pad) {
+  CHECK_CUDA(dy);
+  TORCH_CHECK(dy.dim() == 4 &&
+              dy.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "maxpool_bwd expects NHWC dy");
+  const int Nb = dy.size(0), C = dy.size(1), P = dy.size(2), Q = dy.size(3);
+  auto dx = at::empty({Nb, C, (long)H, (long)W},
+                      dy.options().memory_format(at::MemoryFormat::ChannelsLast));
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::Half, at::ScalarType::BFloat16,
+      dy.scalar_type(), "maxpool_bwd", [&] {
+    const long nvec = (long)Nb * H * W * C / (16 / sizeof(scalar_t));
+    const int blocks = (int)std::min((nvec + 255) / 256, (long)4096);
+    hipLaunchKernelGGL(maxpool_bwd_nhwc_kernel<scalar_t>,
+                       dim3(std::max(blocks, 1)), dim3(256), 0, cur_stream(),
+                       dy.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
+                       dx.data_ptr<scalar_t>(), Nb, (int)H, (int)W, C, P, Q,
+                       (int)k, (int)stride, (int)pad);
+  });
+  return dx;
+}
+
 at::Tensor class_rank(at::Tensor logits, at::Tensor target) {
   CHECK_CUDA(logits); CHECK_CONTIG(logits);
   CHECK_CUDA(target); CHECK_CONTIG(target);
@@ -1399,5 +1549,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "multi-tensor grad unscale + inf/nan check");
   m.def("gap_fwd", &gap_fwd, "global average pool forward (NCHW/NHWC)");
   m.def("gap_bwd", &gap_bwd, "global average pool backward");
+  m.def("maxpool_fwd", &maxpool_fwd,
+        "NHWC max pool forward -> (y, argmax idx)");
+  m.def("maxpool_bwd", &maxpool_bwd, "NHWC max pool backward (gather)");
   m.def("class_rank", &class_rank, "per-row rank of target class (for top-k)");
 }

@@ -55,7 +55,9 @@ def build_training(cfg: TrainConfig, device: torch.device, world_size: int,
     model = build_model(cfg.arch, cfg.num_classes).to(device)
     if cfg.native_ops:
         from ..ops import MI355Conv2d
+        from ..ops.pool import MI355MaxPool2d
         model = MI355Conv2d.convert(model)
+        model = MI355MaxPool2d.convert(model)
     if cfg.channels_last:
         model = model.to(memory_format=torch.channels_last)
     if distributed and world_size > 1 and cfg.sync_bn:

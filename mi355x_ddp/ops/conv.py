@@ -17,7 +17,7 @@ falls back to the static size heuristic inside the kernels.
 from __future__ import annotations
 
 import os
-from typing import Callable, Dict, Sequence, Tuple
+from typing import Callable, Dict, Optional, Sequence, Tuple
 
 import torch
 import torch.nn as nn
@@ -59,17 +59,29 @@ def _native_ok(x: torch.Tensor, weight: torch.Tensor, stride, padding,
 _TUNE_CACHE: Dict[Tuple, int] = {}
 
 
-def _measure_ms(fn: Callable[[], None], iters: int = 4) -> float:
+def _measure_ms(fn: Callable[[], None], iters: int = 4,
+                abort_above_ms: Optional[float] = None) -> float:
+    """Event-timed mean; one probe run first, and if that alone exceeds
+    `abort_above_ms` the candidate is hopeless (e.g. MIOpen immediate mode
+    falling back to a naive kernel at 50+ ms) — skip the refinement runs
+    instead of paying 5x a pathological time."""
     fn()  # allocator warm-up
     torch.cuda.synchronize()
     start = torch.cuda.Event(enable_timing=True)
     stop = torch.cuda.Event(enable_timing=True)
     start.record()
+    fn()
+    stop.record()
+    stop.synchronize()
+    probe = start.elapsed_time(stop)
+    if abort_above_ms is not None and probe > abort_above_ms:
+        return probe
+    start.record()
     for _ in range(iters):
         fn()
     stop.record()
     stop.synchronize()
-    return start.elapsed_time(stop)
+    return start.elapsed_time(stop) / iters
 
 
 def _tuned_choice(key: Tuple, candidates: Sequence,
@@ -86,7 +98,10 @@ def _tuned_choice(key: Tuple, candidates: Sequence,
         return candidates[0] if default is None else default
     best, best_ms = candidates[0], float("inf")
     for c in candidates:
-        ms = _measure_ms(lambda: run(c))
+        # a candidate already 2x slower than the best can't win — one
+        # probe run suffices (bounds the cost of pathological candidates)
+        cutoff = 2.0 * best_ms if best_ms < float("inf") else None
+        ms = _measure_ms(lambda: run(c), abort_above_ms=cutoff)
         if ms < best_ms:
             best, best_ms = c, ms
     _TUNE_CACHE[key] = best

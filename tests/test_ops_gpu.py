@@ -363,3 +363,45 @@ def test_conv_build_wT():
     got = _backend.C().conv_build_wT(w)
     ref = w.flip(2, 3).permute(1, 2, 3, 0).contiguous()
     assert torch.equal(got, ref)
+
+
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float16])
+def test_maxpool_nhwc_parity(dtype):
+    """Native NHWC max pool (fwd argmax + gather backward) vs ATen, including
+    the overlapping-window 3x3 s2 case of the ImageNet stem."""
+    torch.manual_seed(3)
+    for (n, c, h, w, k, s, p) in [(4, 64, 112, 112, 3, 2, 1),
+                                  (2, 64, 56, 56, 2, 2, 0),
+                                  (3, 128, 17, 17, 3, 2, 1)]:
+        x = torch.randn(n, c, h, w, device="cuda", dtype=dtype) \
+            .to(memory_format=torch.channels_last).requires_grad_(True)
+        xr = x.detach().clone().requires_grad_(True)
+        from mi355x_ddp.ops.pool import MI355MaxPool2d
+        pool = MI355MaxPool2d(k, s, p)
+        y = pool(x)
+        yr = torch.nn.functional.max_pool2d(xr, k, s, p)
+        assert torch.equal(y, yr), (n, c, h, w, k, s, p)
+        dy = torch.randn_like(y)
+        y.backward(dy)
+        yr.backward(dy)
+        assert torch.allclose(x.grad.float(), xr.grad.float(),
+                              atol=1e-2, rtol=1e-2), (n, c, h, w, k, s, p)
+
+
+def test_maxpool_in_imagenet_model_step():
+    """resnet50_imagenet forward+backward on the native NHWC path."""
+    from mi355x_ddp.models import build_model
+    from mi355x_ddp.ops import MI355Conv2d
+    from mi355x_ddp.ops.pool import MI355MaxPool2d
+    torch.manual_seed(0)
+    m = build_model("resnet50_imagenet").cuda()
+    m = MI355Conv2d.convert(m)
+    m = MI355MaxPool2d.convert(m)
+    m = m.to(memory_format=torch.channels_last)
+    x = torch.randn(4, 3, 96, 96, device="cuda") \
+        .to(memory_format=torch.channels_last)
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        loss = m(x).float().logsumexp(dim=1).mean()
+    loss.backward()
+    torch.cuda.synchronize()
+    assert torch.isfinite(loss)
