@@ -554,23 +554,26 @@ __device__ __forceinline__ void tr4x4_bf16(const uint2 in[4], uint2 out[4]) {
 }
 
 template <int WGM, int WGN, bool DIRECT>
-__global__ __launch_bounds__(256)
+__global__ __launch_bounds__(256, 2)
 void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
                           const __bf16* __restrict__ x,
                           float* __restrict__ dwp,
                           __bf16* __restrict__ dwb, ConvDims d) {
   constexpr int TM = WGM * 64;          // kout tile
   constexpr int TN = WGN * 64;          // rsc tile
+  constexpr int TA = TM / 64;
+  constexpr int TB = TN / 64;
   __shared__ __bf16 sA[TM * LDK];       // [kout][npq]
   __shared__ __bf16 sB[TN * LDK];       // [rsc][npq]
-  // per-chunk row metadata: for npq row r, the x base offset of tap (0,0),
-  // the (ih0, iw0) coords for bounds tests, and a validity flag (computed
-  // once per chunk by threads 0..63 instead of 4x per staged 4x4 block).
-  // NOTE the flag is separate: the tap-(0,0) offset is legitimately
-  // negative for boundary rows of image 0 (ih0/iw0 = -pad).
-  __shared__ int sRowOff[64];           // (xn*GH + ih0)*GW + iw0
-  __shared__ short sIh0[64], sIw0[64];
-  __shared__ unsigned char sOk[64];     // npq < d.K
+  // DOUBLE-BUFFERED per-chunk row metadata: for npq row r, the x base
+  // offset of tap (0,0), the (ih0, iw0) coords for bounds tests, and a
+  // validity flag (NOT derived from the offset sign — that offset is
+  // legitimately negative for pad-boundary rows of image 0). Buffer b
+  // holds the NEXT chunk's rows so its gathers can issue before the
+  // current chunk's MFMAs (register pipeline, as in the LIN kernel).
+  __shared__ int sRowOff[2][64];        // (xn*GH + ih0)*GW + iw0
+  __shared__ short sIh0[2][64], sIw0[2][64];
+  __shared__ unsigned char sOk[2][64];  // npq < d.K
 
   const int tid = threadIdx.x;
   const int m0 = blockIdx.y * TM;
@@ -587,13 +590,13 @@ void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
   const int nchunks = (d.K + BK - 1) / BK;
   f32x4 acc[4][4] = {};
 
-  for (int kc = blockIdx.z; kc < nchunks; kc += gridDim.z) {
-    const int kk0 = kc * BK;
-    // ---- row metadata (threads 0..63) --------------------------------
+  uint2 sta[TA][4], stb[TB][4];         // staged 4x4 blocks for one chunk
+
+  auto compute_meta = [&](int kc, int b) {
     if (tid < 64) {
-      const int npq = kk0 + tid;
+      const int npq = kc * BK + tid;
       int off = 0, ih0 = 0, iw0 = 0;
-      const bool ok = npq < d.K;
+      const bool ok = npq < d.K && kc < nchunks;
       if (ok) {
         int xn, xp, xq;
         decode_m(npq, d, xn, xp, xq);
@@ -601,98 +604,94 @@ void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
         iw0 = xq * d.stride - d.pad;
         off = ((xn * d.GH + ih0) * d.GW + iw0);
       }
-      sRowOff[tid] = off;
-      sIh0[tid] = (short)ih0;
-      sIw0[tid] = (short)iw0;
-      sOk[tid] = ok;
+      sRowOff[b][tid] = off;
+      sIh0[b][tid] = (short)ih0;
+      sIw0[b][tid] = (short)iw0;
+      sOk[b][tid] = ok;
     }
-    __syncthreads();
-
-    // ---- dy tile: TM x 64, 4x4-register-transposed -------------------
-    // thread -> 4x4 block mapping is COLUMN-group-fastest so a wave's
-    // global loads are contiguous (adjacent lanes read adjacent 8 B of the
-    // same dy row). The transposed b64 stores then land on clashing banks
-    // (4-row stride x 36 dwords = same bank pair), so the npq quad is
-    // XOR-swizzled by the row's 4x4-block index — even-only bits keep
-    // 8-element groups intact for the b128 fragment reads, which apply the
-    // same swizzle.
-    {
-      constexpr int C4N = TM / 4;       // 4-col groups per row quad
-      constexpr int TA = TM / 64;       // staging passes
-      #pragma unroll
-      for (int t = 0; t < TA; ++t) {
-        const int bid = t * 256 + tid;
-        const int c4 = bid & (C4N - 1); // kout 4-col group
-        const int r4 = bid / C4N;       // npq 4-row group (0..15)
-        const int npq0 = kk0 + r4 * 4;
-        const int mcol = m0 + c4 * 4;
-        uint2 in[4], out[4];
-        if (npq0 + 4 <= d.K && mcol + 4 <= d.M) {
-          #pragma unroll
-          for (int i = 0; i < 4; ++i)
-            in[i] = *(const uint2*)(dy + (long)(npq0 + i) * d.M + mcol);
-        } else {
-          #pragma unroll
-          for (int i = 0; i < 4; ++i) {
-            __bf16 e[4] = {};
-            if (npq0 + i < d.K)
-              #pragma unroll
-              for (int c = 0; c < 4; ++c)
-                if (mcol + c < d.M)
-                  e[c] = dy[(long)(npq0 + i) * d.M + mcol + c];
-            in[i] = *(uint2*)e;
-          }
-        }
-        tr4x4_bf16(in, out);
+  };
+  // thread -> 4x4 block mapping is COLUMN-group-fastest so a wave's global
+  // loads are contiguous; the transposed b64 stores are XOR-swizzled on the
+  // npq quad (even bits only — 8-element fragment reads stay intact).
+  auto issue_all = [&](int kk0, int b) {
+    #pragma unroll
+    for (int t = 0; t < TA; ++t) {
+      const int bid = t * 256 + tid;
+      const int c4 = bid & (TM / 4 - 1);
+      const int r4 = bid / (TM / 4);
+      const int npq0 = kk0 + r4 * 4;
+      const int mcol = m0 + c4 * 4;
+      if (npq0 + 4 <= d.K && mcol + 4 <= d.M) {
+        #pragma unroll
+        for (int i = 0; i < 4; ++i)
+          sta[t][i] = *(const uint2*)(dy + (long)(npq0 + i) * d.M + mcol);
+      } else {
         #pragma unroll
         for (int i = 0; i < 4; ++i) {
-          const int row = c4 * 4 + i;
-          const int q = r4 ^ (c4 & 14);
-          *(uint2*)(sA + row * LDK + q * 4) = out[i];
+          __bf16 e[4] = {};
+          if (npq0 + i < d.K)
+            #pragma unroll
+            for (int c = 0; c < 4; ++c)
+              if (mcol + c < d.M)
+                e[c] = dy[(long)(npq0 + i) * d.M + mcol + c];
+          sta[t][i] = *(uint2*)e;
         }
       }
     }
-    // ---- x tile: TN x 64, gather via row metadata --------------------
-    {
-      constexpr int C4N = TN / 4;
-      constexpr int TB = TN / 64;
+    #pragma unroll
+    for (int t = 0; t < TB; ++t) {
+      const int bid = t * 256 + tid;
+      const int c4 = bid & (TN / 4 - 1);
+      const int r4 = bid / (TN / 4);
+      const int nn = n0 + c4 * 4;       // rsc col of this 4-col group
+      // FAST contract (C % 64 == 0): nn..nn+3 sit inside ONE filter tap
+      const int i_tap = nn / SC;
+      const int rem = nn - i_tap * SC;
+      const int j_tap = rem / d.GC;
+      const int cch = rem - j_tap * d.GC;
+      const bool ncol_ok = nn + 4 <= d.N;
       #pragma unroll
-      for (int t = 0; t < TB; ++t) {
-        const int bid = t * 256 + tid;
-        const int c4 = bid & (C4N - 1);
-        const int r4 = bid / C4N;
-        const int nn = n0 + c4 * 4;     // rsc col of this 4-col group
-        // FAST contract (C % 64 == 0): nn..nn+3 sit inside ONE filter tap
-        const int i_tap = nn / SC;
-        const int rem = nn - i_tap * SC;
-        const int j_tap = rem / d.GC;
-        const int cch = rem - j_tap * d.GC;
-        uint2 in[4], out[4];
-        const bool ncol_ok = nn + 4 <= d.N;
-        #pragma unroll
-        for (int i = 0; i < 4; ++i) {
-          const int r = r4 * 4 + i;
-          uint2 v = {0u, 0u};
-          if (ncol_ok && sOk[r]) {
-            const int ih = sIh0[r] + i_tap;
-            const int iw = sIw0[r] + j_tap;
-            if (ih >= 0 && ih < d.GH && iw >= 0 && iw < d.GW)
-              v = *(const uint2*)(x +
-                  ((long)sRowOff[r] + i_tap * d.GW + j_tap) * d.GC + cch);
-          }
-          in[i] = v;
+      for (int i = 0; i < 4; ++i) {
+        const int r = r4 * 4 + i;
+        uint2 v = {0u, 0u};
+        if (ncol_ok && sOk[b][r]) {
+          const int ih = sIh0[b][r] + i_tap;
+          const int iw = sIw0[b][r] + j_tap;
+          if (ih >= 0 && ih < d.GH && iw >= 0 && iw < d.GW)
+            v = *(const uint2*)(x +
+                ((long)sRowOff[b][r] + i_tap * d.GW + j_tap) * d.GC + cch);
         }
-        tr4x4_bf16(in, out);
-        #pragma unroll
-        for (int i = 0; i < 4; ++i) {
-          const int row = c4 * 4 + i;
-          const int q = r4 ^ (c4 & 14);
-          *(uint2*)(sB + row * LDK + q * 4) = out[i];
-        }
+        stb[t][i] = v;
       }
     }
-    __syncthreads();
-
+  };
+  auto write_all = [&]() {
+    #pragma unroll
+    for (int t = 0; t < TA; ++t) {
+      const int bid = t * 256 + tid;
+      const int c4 = bid & (TM / 4 - 1);
+      const int r4 = bid / (TM / 4);
+      uint2 out[4];
+      tr4x4_bf16(sta[t], out);
+      const int q = r4 ^ (c4 & 14);
+      #pragma unroll
+      for (int i = 0; i < 4; ++i)
+        *(uint2*)(sA + (c4 * 4 + i) * LDK + q * 4) = out[i];
+    }
+    #pragma unroll
+    for (int t = 0; t < TB; ++t) {
+      const int bid = t * 256 + tid;
+      const int c4 = bid & (TN / 4 - 1);
+      const int r4 = bid / (TN / 4);
+      uint2 out[4];
+      tr4x4_bf16(stb[t], out);
+      const int q = r4 ^ (c4 & 14);
+      #pragma unroll
+      for (int i = 0; i < 4; ++i)
+        *(uint2*)(sB + (c4 * 4 + i) * LDK + q * 4) = out[i];
+    }
+  };
+  auto mfma_all = [&]() {
     #pragma unroll
     for (int ks = 0; ks < BK; ks += 32) {
       bf16x8 af[4], bf[4];
@@ -715,7 +714,35 @@ void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
     }
-    __syncthreads();
+  };
+
+  // pipeline: meta(k) -> loads(k) -> tiles(k); then per chunk k:
+  //   loads(k+1) [covered by MFMA(k)] / MFMA(k) / write tiles(k+1) +
+  //   meta(k+2) between two barriers
+  bool first = true;
+  int buf = 0;
+  for (int kc = blockIdx.z; kc < nchunks; kc += gridDim.z) {
+    const int kn = kc + gridDim.z;
+    if (first) {
+      compute_meta(kc, 0);
+      __syncthreads();
+      issue_all(kc * BK, 0);
+      write_all();
+      compute_meta(kn, 1);
+      __syncthreads();
+      first = false;
+      buf = 1;                          // meta[1] holds chunk kn
+    }
+    const bool has_next = kn < nchunks;
+    if (has_next) issue_all(kn * BK, buf);
+    mfma_all();
+    if (has_next) {
+      __syncthreads();
+      write_all();
+      compute_meta(kn + gridDim.z, buf ^ 1);
+      __syncthreads();
+      buf ^= 1;
+    }
   }
 
   const int dm = (lane >> 4) * 4;
