@@ -26,8 +26,10 @@ import torch.distributed as dist
 # bf16/fp16 runs against the DDP+apex AMP row (14.5 s/epoch = 3448.3 img/s),
 # reference README.md:76-77.
 REFERENCE_IMAGES_PER_SEC = {"fp32": 3125.0, "bf16": 50000.0 / 14.5,
-                            "fp16": 50000.0 / 14.5}
-REFERENCE_SECONDS_PER_EPOCH = {"fp32": 16.0, "bf16": 14.5, "fp16": 14.5}
+                            "fp16": 50000.0 / 14.5,
+                            "bf16_o2": 50000.0 / 14.5}
+REFERENCE_SECONDS_PER_EPOCH = {"fp32": 16.0, "bf16": 14.5, "fp16": 14.5,
+                               "bf16_o2": 14.5}
 EPOCH_IMAGES = 50000  # CIFAR-100 train set size
 
 
@@ -38,7 +40,8 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--global-batch", "--global_batch", type=int, default=256)
     p.add_argument("--arch", type=str, default="resnet18")
-    p.add_argument("--amp", type=str, default="bf16", choices=["fp32", "bf16", "fp16"])
+    p.add_argument("--amp", type=str, default="bf16",
+                   choices=["fp32", "bf16", "fp16", "bf16_o2"])
     p.add_argument("--mode", type=str, default="flat", choices=["flat", "torchddp"])
     p.add_argument("--grad-accu-steps", type=int, default=1)
     p.add_argument("--no-syncbn", action="store_true")

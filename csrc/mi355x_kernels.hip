@@ -702,6 +702,50 @@ __global__ void multi_tensor_sgd_kernel(MTTensorList tl, float lr, float mu,
 }
 
 // ---------------------------------------------------------------------------
+// Mixed-precision multi-tensor SGD (apex O2 equivalent): bf16 model
+// weights + bf16 grads + fp32 MASTER weights and momentum. The update runs
+// in fp32 against the master; the bf16 parameter is the rounded copy.
+//   m <- mu*m + g + wd*master ; master <- master - lr*m ; p <- (bf16)master
+// ---------------------------------------------------------------------------
+struct MTMixedList {
+  __bf16* p[MT_MAX_TENSORS];
+  const __bf16* g[MT_MAX_TENSORS];
+  float* w[MT_MAX_TENSORS];   // master
+  float* m[MT_MAX_TENSORS];
+  int size[MT_MAX_TENSORS];
+  int ntensors;
+};
+
+__global__ void multi_tensor_sgd_o2_kernel(MTMixedList tl, float lr, float mu,
+                                           float wd, bool has_momentum) {
+  int b = blockIdx.x;
+  int t = 0;
+  int chunk = 0;
+  for (; t < tl.ntensors; ++t) {
+    const int nchunks = (tl.size[t] + MT_CHUNK - 1) / MT_CHUNK;
+    if (b < nchunks) { chunk = b; break; }
+    b -= nchunks;
+  }
+  if (t >= tl.ntensors) return;
+  const int start = chunk * MT_CHUNK;
+  const int end = min(start + MT_CHUNK, tl.size[t]);
+  __bf16* p = tl.p[t];
+  const __bf16* g = tl.g[t];
+  float* w = tl.w[t];
+  float* m = has_momentum ? tl.m[t] : nullptr;
+  for (int i = start + threadIdx.x; i < end; i += blockDim.x) {
+    float grad = (float)g[i] + wd * w[i];
+    if (has_momentum) {
+      grad = mu * m[i] + grad;
+      m[i] = grad;
+    }
+    const float nw = w[i] - lr * grad;
+    w[i] = nw;
+    p[i] = (__bf16)nw;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Multi-tensor unscale + inf/nan check (fp16 loss-scaler backward pass).
 // ---------------------------------------------------------------------------
 struct MTGradList {
@@ -1368,6 +1412,45 @@ void multi_tensor_sgd(std::vector<at::Tensor> params,
   }
 }
 
+void multi_tensor_sgd_o2(std::vector<at::Tensor> params,
+                         std::vector<at::Tensor> grads,
+                         std::vector<at::Tensor> masters,
+                         std::vector<at::Tensor> bufs,
+                         double lr, double momentum, double weight_decay) {
+  const bool has_momentum = !bufs.empty();
+  TORCH_CHECK(params.size() == grads.size() &&
+              params.size() == masters.size());
+  auto stream = cur_stream();
+  size_t i = 0;
+  while (i < params.size()) {
+    MTMixedList tl;
+    int nblocks = 0;
+    int t = 0;
+    for (; t < MT_MAX_TENSORS && i < params.size(); ++t, ++i) {
+      auto& p = params[i];
+      TORCH_CHECK(p.is_cuda() && p.is_non_overlapping_and_dense() &&
+                  p.scalar_type() == at::ScalarType::BFloat16,
+                  "multi_tensor_sgd_o2 expects dense bf16 params");
+      TORCH_CHECK(grads[i].strides() == p.strides() &&
+                  grads[i].scalar_type() == at::ScalarType::BFloat16,
+                  "multi_tensor_sgd_o2: grad layout/dtype mismatch");
+      TORCH_CHECK(masters[i].strides() == p.strides() &&
+                  masters[i].scalar_type() == at::kFloat,
+                  "multi_tensor_sgd_o2: master layout/dtype mismatch");
+      tl.p[t] = reinterpret_cast<__bf16*>(p.data_ptr());
+      tl.g[t] = reinterpret_cast<const __bf16*>(grads[i].data_ptr());
+      tl.w[t] = masters[i].data_ptr<float>();
+      tl.m[t] = has_momentum ? bufs[i].data_ptr<float>() : nullptr;
+      tl.size[t] = (int)p.numel();
+      nblocks += (tl.size[t] + MT_CHUNK - 1) / MT_CHUNK;
+    }
+    tl.ntensors = t;
+    hipLaunchKernelGGL(multi_tensor_sgd_o2_kernel, dim3(nblocks), dim3(256),
+                       0, stream, tl, (float)lr, (float)momentum,
+                       (float)weight_decay, has_momentum);
+  }
+}
+
 at::Tensor multi_tensor_unscale(std::vector<at::Tensor> grads, double inv_scale) {
   TORCH_CHECK(!grads.empty());
   auto found = at::zeros({1}, grads[0].options().dtype(at::kInt));
@@ -1545,6 +1628,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("xent_fwd", &xent_fwd, "fused softmax cross-entropy forward");
   m.def("xent_bwd", &xent_bwd, "fused softmax cross-entropy backward");
   m.def("multi_tensor_sgd", &multi_tensor_sgd, "fused multi-tensor SGD step");
+  m.def("multi_tensor_sgd_o2", &multi_tensor_sgd_o2,
+        "fused SGD with bf16 params/grads + fp32 master (apex O2)");
   m.def("multi_tensor_unscale", &multi_tensor_unscale,
         "multi-tensor grad unscale + inf/nan check");
   m.def("gap_fwd", &gap_fwd, "global average pool forward (NCHW/NHWC)");

@@ -35,7 +35,9 @@ class TrainConfig:
     lr_gamma: float = 0.2
     grad_accu_steps: int = 1          # micro-batching; collectives elided on non-final micro-steps
 
-    # precision: "fp32" | "bf16" | "fp16" (fp16 uses the dynamic loss scaler)
+    # precision: "fp32" | "bf16" | "fp16" | "bf16_o2"
+    # (fp16 uses the dynamic loss scaler; bf16_o2 = apex-O2-equivalent pure
+    # bf16 model with fp32 master weights in the optimizer)
     amp: str = "fp32"
 
     # distributed
@@ -93,7 +95,8 @@ def add_common_args(p: argparse.ArgumentParser) -> argparse.ArgumentParser:
                    choices=["resnet18", "resnet34", "resnet50", "resnet101",
                             "resnet152", "resnet18_imagenet",
                             "resnet50_imagenet"])
-    p.add_argument("--amp", default=None, type=str, choices=[None, "fp32", "bf16", "fp16"])
+    p.add_argument("--amp", default=None, type=str,
+                   choices=[None, "fp32", "bf16", "fp16", "bf16_o2"])
     p.add_argument("--no-sync-bn", action="store_true")
     p.add_argument("--num_workers", default=4, type=int)
     p.add_argument("--log_interval", default=10, type=int)

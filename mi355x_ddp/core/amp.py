@@ -15,7 +15,11 @@ import torch
 
 from ..ops import _backend
 
-_DTYPES = {"bf16": torch.bfloat16, "fp16": torch.float16}
+# "bf16_o2" = apex-O2 equivalent: the MODEL's parameters are bf16 (fp32
+# masters live in FusedSGD state), so autocast's per-forward weight casts
+# disappear; activations run bf16 like plain "bf16"
+_DTYPES = {"bf16": torch.bfloat16, "fp16": torch.float16,
+           "bf16_o2": torch.bfloat16}
 
 
 def autocast_ctx(mode: str, device_type: Optional[str] = None):
@@ -93,3 +97,12 @@ class DynamicLossScaler:
 
 def build_scaler(mode: str) -> Optional[DynamicLossScaler]:
     return DynamicLossScaler() if mode == "fp16" else None
+
+
+def cast_model_bf16(model):
+    """apex-O2 model prep: every parameter re-stored as bf16 (buffers —
+    BN running stats — stay fp32). FusedSGD creates the fp32 masters at
+    first step."""
+    for p in model.parameters():
+        p.data = p.data.to(torch.bfloat16)
+    return model

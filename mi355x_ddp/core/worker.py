@@ -62,9 +62,16 @@ def build_training(cfg: TrainConfig, device: torch.device, world_size: int,
         model = model.to(memory_format=torch.channels_last)
     if distributed and world_size > 1 and cfg.sync_bn:
         model = MI355SyncBatchNorm.convert_sync_batchnorm(model)
+    if cfg.amp == "bf16_o2":
+        from .amp import cast_model_bf16
+        model = cast_model_bf16(model)
     if distributed and world_size >= 1 and wrap == "flat":
         model = FlatDDP(model, bucket_cap_mb=cfg.bucket_cap_mb,
                         overlap=not cfg.hip_graph,
+                        # O2: the whole model (and its grads) is bf16, so the
+                        # flat buffer and the wire are bf16 natively
+                        grad_dtype=torch.bfloat16 if cfg.amp == "bf16_o2"
+                        else torch.float32,
                         comm_dtype=torch.bfloat16 if cfg.comm_bf16 else None,
                         # hipGraph capture needs static grad memory even at
                         # world 1; otherwise world 1 skips the flat buffer
