@@ -331,5 +331,9 @@ def test_o2_sgd_kernel_matches_fallback():
     torch.cuda.synchronize()
     for a, b in zip(wn, wf):
         assert torch.allclose(a, b, atol=1e-6), (a - b).abs().max()
-    for a, b in zip(pn, pf):
-        assert torch.equal(a, b)
+    # the kernel fuses mu*m+g (fma) where the eager fallback does mul+add,
+    # so masters can differ in the last ulp and round to different bf16 at
+    # exact ties — the invariant is each path's param == its OWN rounded
+    # master, and masters near-identical (above)
+    for pp, ww in list(zip(pn, wn)) + list(zip(pf, wf)):
+        assert torch.equal(pp, ww.bfloat16())
