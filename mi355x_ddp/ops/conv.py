@@ -94,7 +94,11 @@ def _tuned_choice(key: Tuple, candidates: Sequence,
     """
     if key in _TUNE_CACHE:
         return _TUNE_CACHE[key]
-    if not autotune_wanted() or torch.cuda.is_current_stream_capturing():
+    # NB: is_current_stream_capturing() raises on CPU-only machines — check
+    # the cheap conditions first (the tuner is only reached from GPU paths,
+    # but keep it robust for direct callers/tests)
+    if (not autotune_wanted() or not torch.cuda.is_available() or
+            torch.cuda.is_current_stream_capturing()):
         return candidates[0] if default is None else default
     best, best_ms = candidates[0], float("inf")
     for c in candidates:

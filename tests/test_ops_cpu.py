@@ -214,3 +214,57 @@ def test_o2_mode_trains_and_tracks_fp32():
     lfp = run("fp32")
     for i, (a, b) in enumerate(zip(lo2, lfp)):
         assert abs(a - b) < 0.05 + 0.05 * i, (lo2, lfp)
+
+
+def test_autotune_cache_semantics(monkeypatch):
+    """The per-shape cache measures once, caches the winner, and returns the
+    declared default (never MIOpen) when autotune is off."""
+    from mi355x_ddp.ops import conv as convmod
+    convmod.clear_autotune_cache()
+    calls = []
+
+    def fake_measure(fn, iters=4, abort_above_ms=None):
+        fn()
+        return {"a": 3.0, "b": 1.0, "c": 2.0}[fake_measure.current]
+
+    def run(c):
+        fake_measure.current = c
+        calls.append(c)
+
+    monkeypatch.setattr(convmod, "_measure_ms",
+                        lambda fn, **kw: (fn(), {"a": 3.0, "b": 1.0,
+                                                 "c": 2.0}[calls[-1]])[1])
+    # exercise the measuring path without a GPU
+    monkeypatch.setattr(convmod.torch.cuda, "is_available", lambda: True)
+    monkeypatch.setattr(convmod.torch.cuda, "is_current_stream_capturing",
+                        lambda: False)
+    key = ("test", 1, 2, 3)
+    got = convmod._tuned_choice(key, ("a", "b", "c"), run, default="a")
+    assert got == "b"
+    n_measured = len(calls)
+    calls.clear()
+    # second lookup: cached, zero measurements
+    assert convmod._tuned_choice(key, ("a", "b", "c"), run, default="a") == "b"
+    assert not calls and n_measured == 3
+
+    # autotune off: default, no measurement, no cache pollution
+    monkeypatch.setenv("MI355X_AUTOTUNE", "0")
+    key2 = ("test2",)
+    assert convmod._tuned_choice(key2, ("a", "b"), run, default="zz") == "zz"
+    assert key2 not in convmod._TUNE_CACHE
+    convmod.clear_autotune_cache()
+
+
+def test_bucket_cap_policy():
+    """World-size-keyed bucket policy: no comm at world 1; 2 buckets for
+    ResNet-sized models; env override wins."""
+    from mi355x_ddp.parallel.flat_ddp import bucket_cap_for
+    assert bucket_cap_for(1, 45.0) == 45.0          # one nominal bucket
+    assert bucket_cap_for(8, 45.0) == 22.5          # 2 buckets
+    assert bucket_cap_for(8, 400.0) == 100.0        # ~4 buckets, big model
+    import os
+    os.environ["MI355X_BUCKET_CAP_MB"] = "7.5"
+    try:
+        assert bucket_cap_for(8, 45.0) == 7.5
+    finally:
+        del os.environ["MI355X_BUCKET_CAP_MB"]
