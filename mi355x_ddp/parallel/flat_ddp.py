@@ -95,7 +95,10 @@ class FlatDDP(nn.Module):
         self._params = params
         numel = sum(p.numel() for p in params)
         device = params[0].device if params else torch.device("cpu")
-        total_mb = numel * 4 / (1024 * 1024)
+        # size the buckets by the BYTES the wire moves (bf16 grad buffers
+        # halve them), not a hard-coded fp32 assumption
+        esize = torch.tensor([], dtype=grad_dtype).element_size()
+        total_mb = numel * esize / (1024 * 1024)
         if bucket_cap_mb is None:
             bucket_cap_mb = bucket_cap_for(self.world_size, total_mb)
 
