@@ -29,6 +29,7 @@ CONV_SHAPES = [
     (3, 64, 32, 32, 64, 3, 1, 1),      # ragged batch -> M % 128 != 0
     (5, 3, 30, 30, 64, 3, 1, 1),       # ragged stem (PADC path, M % 64 != 0)
     (8, 3, 224, 224, 64, 3, 1, 1),     # ImageNet-shape stem (PADC, big M)
+    (4, 3, 64, 64, 64, 7, 2, 3),       # 7x7 s2 ImageNet stem (PADC, K=196->256)
 ]
 
 
