@@ -151,8 +151,12 @@ def main():
             scaler.step(optimizer)
         else:
             optimizer.step()
+        flush_nbt()
 
     # --- warmup ------------------------------------------------------------
+    from mi355x_ddp.ops.batchnorm import (
+        defer_num_batches_tracked, flush_num_batches_tracked as flush_nbt)
+    defer_num_batches_tracked(True)
     model.train()
     import sys
     for i in range(args.warmup):

@@ -17,6 +17,8 @@ from typing import Optional
 import torch
 
 from ..config import TrainConfig
+from ..ops.batchnorm import (defer_num_batches_tracked,
+                             flush_num_batches_tracked)
 from . import dist as dist_utils
 from .amp import DynamicLossScaler, autocast_ctx
 from .metrics import AverageMeter, JsonlSink, ProgressMeter, accuracy
@@ -39,6 +41,7 @@ def train_one_epoch(model, loader, criterion, optimizer, epoch: int,
                     cfg: TrainConfig, device, scaler: Optional[DynamicLossScaler] = None,
                     sink: Optional[JsonlSink] = None, max_steps: Optional[int] = None) -> float:
     model.train()
+    defer_num_batches_tracked(True)
     nprocs = dist_utils.get_world_size()
     accu = max(1, cfg.grad_accu_steps)
     losses = AverageMeter("Loss", ":.4e")
@@ -93,6 +96,7 @@ def train_one_epoch(model, loader, criterion, optimizer, epoch: int,
             scaler.step(optimizer)
         else:
             optimizer.step()
+        flush_num_batches_tracked()
 
         epoch_loss_sum += log_loss
         epoch_steps += 1
@@ -112,6 +116,10 @@ def train_one_epoch(model, loader, criterion, optimizer, epoch: int,
                 if sink is not None:
                     sink.log(kind="train", epoch=epoch, step=step,
                              loss=float(reduced), lr=lr, step_time=step_time.val)
+    # leave no deferral active outside the loop (direct bn_relu callers,
+    # eval paths) and no pending bumps unapplied on early exits
+    flush_num_batches_tracked()
+    defer_num_batches_tracked(False)
     if epoch_steps == 0:
         return 0.0
     mean = dist_utils.reduce_mean(epoch_loss_sum / epoch_steps, nprocs)

@@ -178,10 +178,35 @@ def _group_for(bn: nn.Module):
     return None
 
 
+# Deferred num_batches_tracked updates: every BN otherwise launches a 4 us
+# int64 add kernel per step just to bump its counter (~20 launches on
+# ResNet18). With deferral on, forwards collect the counter buffers and
+# flush_num_batches_tracked() bumps them all in ONE torch._foreach_add_
+# per step. Modules with momentum=None (cumulative averaging READS the
+# counter each forward) keep the eager update.
+_NBT_DEFER = False
+_NBT_PENDING = []
+
+
+def defer_num_batches_tracked(enable: bool = True) -> None:
+    global _NBT_DEFER
+    _NBT_DEFER = enable
+
+
+def flush_num_batches_tracked() -> None:
+    # apply the deferred counter bumps (call once per optimizer step)
+    if _NBT_PENDING:
+        torch._foreach_add_(_NBT_PENDING, 1)
+        _NBT_PENDING.clear()
+
+
 def _run_fused(x, residual, bn: nn.Module, relu: bool):
     training = bn.training or not bn.track_running_stats
     if bn.training and bn.track_running_stats and bn.num_batches_tracked is not None:
-        bn.num_batches_tracked.add_(1)
+        if _NBT_DEFER and bn.momentum is not None:
+            _NBT_PENDING.append(bn.num_batches_tracked)
+        else:
+            bn.num_batches_tracked.add_(1)
     momentum = bn.momentum if bn.momentum is not None else (
         1.0 / float(bn.num_batches_tracked.item()) if bn.num_batches_tracked is not None else 0.1)
     return _FusedBNFunction.apply(

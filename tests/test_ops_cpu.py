@@ -268,3 +268,28 @@ def test_bucket_cap_policy():
         assert bucket_cap_for(8, 45.0) == 7.5
     finally:
         del os.environ["MI355X_BUCKET_CAP_MB"]
+
+
+def test_deferred_num_batches_tracked():
+    """Deferred counter bumps: one foreach add applies every BN's pending
+    increment; eager semantics (momentum=None) unaffected."""
+    import torch.nn as nn
+    from mi355x_ddp.ops import bn_relu
+    from mi355x_ddp.ops.batchnorm import (defer_num_batches_tracked,
+                                          flush_num_batches_tracked)
+    bns = [nn.BatchNorm2d(8) for _ in range(3)]
+    x = torch.randn(2, 8, 4, 4)
+    defer_num_batches_tracked(True)
+    try:
+        for bn in bns:
+            bn_relu(x, bn)
+        assert all(int(bn.num_batches_tracked) == 0 for bn in bns)
+        flush_num_batches_tracked()
+        assert all(int(bn.num_batches_tracked) == 1 for bn in bns)
+        # momentum=None reads the counter inside the forward: stays eager
+        bn_none = nn.BatchNorm2d(8, momentum=None)
+        bn_relu(x, bn_none)
+        assert int(bn_none.num_batches_tracked) == 1
+    finally:
+        defer_num_batches_tracked(False)
+        flush_num_batches_tracked()
