@@ -25,8 +25,13 @@ MATRIX = [
     ("fp16 + loss scaler", ["--amp", "fp16"]),
     ("bf16 + grad_accu 4", ["--grad-accu-steps", "4"]),
     ("bf16, torch-DDP reducer A/B", ["--mode", "torchddp"]),
+    ("bf16_o2 (pure-bf16 model, fp32 masters)", ["--amp", "bf16_o2"]),
+    ("bf16 + bf16 gradient comm", ["--comm-bf16"]),
     ("ResNet50", ["--arch", "resnet50"]),
     ("ResNet34", ["--arch", "resnet34"]),
+    ("ResNet50-ImageNet @224 b64",
+     ["--arch", "resnet50_imagenet", "--image-size", "224",
+      "--global-batch", "64"]),
 ]
 
 
