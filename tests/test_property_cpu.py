@@ -91,3 +91,61 @@ def test_fused_sgd_matches_torch_sgd_fuzz(nt, momentum, wd, steps):
         ref.step()
     for p, q in zip(ps, qs):
         assert torch.allclose(p, q, atol=1e-6)
+
+
+@given(nt=st.integers(1, 4), momentum=st.sampled_from([0.0, 0.9]),
+       wd=st.sampled_from([0.0, 1e-2]), steps=st.integers(1, 3))
+def test_fused_sgd_o2_master_matches_fp32_sgd_fuzz(nt, momentum, wd, steps):
+    """O2 (bf16 params + fp32 masters): the MASTER trajectory equals plain
+    fp32 SGD fed the same (bf16-rounded) gradients; the bf16 params are the
+    rounded masters."""
+    from mi355x_ddp.ops import FusedSGD
+    g = torch.Generator().manual_seed(nt * 13 + int(momentum * 10) + steps)
+    shapes = [(4, 6), (9,), (2, 2, 3), (5,)][:nt]
+    w0 = [torch.randn(*s, generator=g) for s in shapes]
+    ps = [w.bfloat16() for w in w0]
+    opt = FusedSGD(ps, lr=0.1, momentum=momentum, weight_decay=wd)
+    refs = [p.float().requires_grad_(True) for p in ps]
+    ref = torch.optim.SGD(refs, lr=0.1, momentum=momentum, weight_decay=wd)
+    for i in range(steps):
+        gen = torch.Generator().manual_seed(300 + i)
+        grads = [torch.randn(*s, generator=gen).bfloat16() for s in shapes]
+        for p, q, gr in zip(ps, refs, grads):
+            p.grad = gr.clone()
+            q.grad = gr.float()
+        opt.step()
+        ref.step()
+    for p, q in zip(ps, refs):
+        assert torch.allclose(opt.state[p]["master"], q.detach(), atol=1e-6)
+        assert torch.equal(p, q.detach().bfloat16())
+
+
+@given(seed=st.integers(0, 50), cap_kb=st.sampled_from([1, 8, 64, 1024]))
+def test_flat_ddp_bucket_tiling_fuzz(seed, cap_kb):
+    """Bucket invariants hold for arbitrary models: buckets tile the flat
+    buffer contiguously/disjointly/completely and every param maps into
+    exactly the bucket containing its span."""
+    import torch.nn as nn
+    from mi355x_ddp.parallel import FlatDDP
+    g = torch.Generator().manual_seed(seed)
+    layers = []
+    dims = [int(torch.randint(1, 40, (1,), generator=g))
+            for _ in range(int(torch.randint(2, 6, (1,), generator=g)))]
+    prev = 7
+    for d in dims:
+        layers += [nn.Linear(prev, d, bias=bool(d % 2))]
+        prev = d
+    model = FlatDDP(nn.Sequential(*layers), bucket_cap_mb=cap_kb / 1024.0,
+                    static_grads=True)
+    numel = sum(p.numel() for p in model._params)
+    assert model.flat_grads.numel() == numel
+    prev_end = 0
+    for (s_, e_, ps) in model._buckets:
+        assert s_ == prev_end and e_ > s_
+        prev_end = e_
+    assert prev_end == numel
+    for p in model._params:
+        off, view = model._views[p]
+        assert p.grad is view and view.shape == p.shape
+        s_, e_, ps = model._buckets[model._param_bucket[p]]
+        assert s_ <= off and off + p.numel() <= e_ and p in ps
