@@ -5,8 +5,10 @@ scaler — and `--amp fp16` for apex-behavioral parity (dynamic loss scaling wit
 a HIP multi-tensor unscale/inf-check kernel; reference used apex.amp at
 distributed_apex.py:86,119-120).
 
-Run: python distributed_apex.py --batch_size 256   # bf16
+Run: python distributed_apex.py --batch_size 256   # bf16 (apex O1 equivalent)
      python distributed_apex.py --amp fp16         # apex-parity loss scaling
+     python distributed_apex.py --amp bf16_o2      # apex O2: bf16 model +
+                                                   # fp32 master weights
 """
 import argparse
 
