@@ -50,6 +50,9 @@ class TrainConfig:
     bucket_cap_mb: Optional[float] = None
     comm_bf16: bool = False           # all-reduce gradients in bf16 (half the wire bytes)
     dist_timeout_s: int = 600         # collective timeout (a dead rank fails fast)
+    # stall diagnostics: dump all-thread stacks to stderr if one training
+    # step exceeds this many seconds (None = off); see core/watchdog.py
+    stall_dump_s: Optional[float] = 300.0
     use_flat_ddp: bool = True         # MI355X-native flat-bucket reducer (graph-capturable)
 
     # data

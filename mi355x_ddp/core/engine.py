@@ -22,6 +22,7 @@ from ..ops.batchnorm import (defer_num_batches_tracked,
 from . import dist as dist_utils
 from .amp import DynamicLossScaler, autocast_ctx
 from .metrics import AverageMeter, JsonlSink, ProgressMeter, accuracy
+from .watchdog import StepWatchdog
 
 
 def _grad_tensors(model):
@@ -51,11 +52,13 @@ def train_one_epoch(model, loader, criterion, optimizer, epoch: int,
     # log_interval subsampling bias in the return value
     epoch_loss_sum = torch.zeros((), device=device)
     epoch_steps = 0
+    watchdog = StepWatchdog(getattr(cfg, "stall_dump_s", None))
     end = time.time()
 
     for step, (images, labels) in enumerate(loader):
         if max_steps is not None and step >= max_steps:
             break
+        watchdog.arm()
         images = images.to(device, non_blocking=True)
         labels = labels.to(device, non_blocking=True)
         if cfg.channels_last:
@@ -97,6 +100,7 @@ def train_one_epoch(model, loader, criterion, optimizer, epoch: int,
         else:
             optimizer.step()
         flush_num_batches_tracked()
+        watchdog.disarm()
 
         epoch_loss_sum += log_loss
         epoch_steps += 1

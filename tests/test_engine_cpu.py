@@ -190,3 +190,31 @@ def test_same_seed_reproduces_losses():
         return losses
 
     assert run() == run()
+
+
+def test_step_watchdog_dumps_on_stall():
+    """A step exceeding the threshold dumps all-thread stacks to stderr;
+    fast steps dump nothing (SURVEY 5.3 failure diagnostics)."""
+    import subprocess
+    import sys
+    import textwrap
+    code = textwrap.dedent("""
+        import time
+        from mi355x_ddp.core.watchdog import StepWatchdog
+        wd = StepWatchdog(0.3)
+        with wd:
+            time.sleep(1.0)    # stalled 'step'
+        with StepWatchdog(5.0):
+            time.sleep(0.01)   # fast step
+        print("done")
+    """)
+    r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=60)
+    assert r.returncode == 0 and "done" in r.stdout
+    assert "Timeout" in r.stderr and r.stderr.count("Timeout") == 1
+
+
+def test_watchdog_none_is_noop():
+    from mi355x_ddp.core.watchdog import StepWatchdog
+    with StepWatchdog(None):
+        pass  # must not arm faulthandler at all
