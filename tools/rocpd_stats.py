@@ -59,18 +59,20 @@ def pmc_stats(path, top=30):
     agg = collections.defaultdict(dict)
     for name, pid, val, cnt in db.execute(q):
         agg[name][pmc_names[pid]] = val
-    rows = []
-    for name, d in agg.items():
-        wc = d.get("SQ_WAVE_CYCLES", 0) or 1
-        rows.append((name, d.get("SQ_VALU_MFMA_BUSY_CYCLES", 0),
-                     d.get("SQ_WAIT_ANY", 0), wc))
-    rows.sort(key=lambda r: -r[3])
-    out = [f"{'mfma%':>7} {'wait%':>7} {'wave_Mcyc':>10}  kernel",
-           "(mfma% = SQ_VALU_MFMA_BUSY_CYCLES/quad / SQ_WAVE_CYCLES;"
-           " wait% = SQ_WAIT_ANY / SQ_WAVE_CYCLES)"]
-    for name, mfma, wait, wc in rows[:top]:
-        out.append(f"{100*(mfma/4)/wc:7.2f} {100*wait/wc:7.2f} "
-                   f"{wc/1e6:10.1f}  {name[:120]}")
+    # report EVERY captured counter (earlier versions hard-coded two and
+    # silently dropped e.g. SQ_LDS_BANK_CONFLICT from the output)
+    counters = sorted({c for d in agg.values() for c in d})
+    ordered = [c for c in ("SQ_WAVE_CYCLES",) if c in counters] + \
+        [c for c in counters if c != "SQ_WAVE_CYCLES"]
+    rows = sorted(agg.items(), key=lambda kv: -kv[1].get("SQ_WAVE_CYCLES", 0))
+    hdr = " ".join(f"{c[:14]:>15}" for c in ordered)
+    out = [f"{hdr}  kernel",
+           "(raw counter sums per kernel; SQ_* are quad-cycles except "
+           "SQ_VALU_MFMA_BUSY_CYCLES which counts cycles; derive ratios "
+           "against SQ_WAVE_CYCLES)"]
+    for name, d in rows[:top]:
+        vals = " ".join(f"{d.get(c, 0):15.3e}" for c in ordered)
+        out.append(f"{vals}  {name[:110]}")
     return "\n".join(out)
 
 
