@@ -553,7 +553,12 @@ __device__ __forceinline__ void tr4x4_bf16(const uint2 in[4], uint2 out[4]) {
   out[3].y = __builtin_amdgcn_perm(in[3].y, in[2].y, HI);
 }
 
-template <int WGM, int WGN, bool DIRECT>
+// CH = BK-chunks per pipeline stage. CH=2 stages 128 npq rows so the MFMA
+// phase (~64 MFMAs/wave, ~1100 cyc) covers the ~900-cyc load latency that
+// left CH=1 at 54-57% SQ_WAIT_ANY (profiles/r2c13): LDS doubles (69.6 KB
+// for (2,2) -> 2 blocks/CU) and staging registers double, trading
+// block-level overlap for complete within-wave latency cover.
+template <int WGM, int WGN, bool DIRECT, int CH = 1>
 __global__ __launch_bounds__(256, 2)
 void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
                           const __bf16* __restrict__ x,
@@ -563,17 +568,19 @@ void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
   constexpr int TN = WGN * 64;          // rsc tile
   constexpr int TA = TM / 64;
   constexpr int TB = TN / 64;
-  __shared__ __bf16 sA[TM * LDK];       // [kout][npq]
-  __shared__ __bf16 sB[TN * LDK];       // [rsc][npq]
+  constexpr int SK = CH * BK;           // npq rows per stage
+  constexpr int SLDK = SK + 8;          // padded LDS row length
+  __shared__ __bf16 sA[TM * SLDK];      // [kout][npq]
+  __shared__ __bf16 sB[TN * SLDK];      // [rsc][npq]
   // DOUBLE-BUFFERED per-chunk row metadata: for npq row r, the x base
   // offset of tap (0,0), the (ih0, iw0) coords for bounds tests, and a
   // validity flag (NOT derived from the offset sign — that offset is
   // legitimately negative for pad-boundary rows of image 0). Buffer b
   // holds the NEXT chunk's rows so its gathers can issue before the
   // current chunk's MFMAs (register pipeline, as in the LIN kernel).
-  __shared__ int sRowOff[2][64];        // (xn*GH + ih0)*GW + iw0
-  __shared__ short sIh0[2][64], sIw0[2][64];
-  __shared__ unsigned char sOk[2][64];  // npq < d.K
+  __shared__ int sRowOff[2][SK];        // (xn*GH + ih0)*GW + iw0
+  __shared__ short sIh0[2][SK], sIw0[2][SK];
+  __shared__ unsigned char sOk[2][SK];  // npq < d.K
 
   const int tid = threadIdx.x;
   const int m0 = blockIdx.y * TM;
@@ -587,16 +594,16 @@ void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
   const int fk = (lane >> 4) * 8;
 
   const int SC = d.S * d.GC;
-  const int nchunks = (d.K + BK - 1) / BK;
+  const int nstages = (d.K + SK - 1) / SK;
   f32x4 acc[4][4] = {};
 
-  uint2 sta[TA][4], stb[TB][4];         // staged 4x4 blocks for one chunk
+  uint2 sta[TA * CH][4], stb[TB * CH][4];  // staged 4x4 blocks for one stage
 
   auto compute_meta = [&](int kc, int b) {
-    if (tid < 64) {
-      const int npq = kc * BK + tid;
+    if (tid < SK) {
+      const int npq = kc * SK + tid;
       int off = 0, ih0 = 0, iw0 = 0;
-      const bool ok = npq < d.K && kc < nchunks;
+      const bool ok = npq < d.K && kc < nstages;
       if (ok) {
         int xn, xp, xq;
         decode_m(npq, d, xn, xp, xq);
@@ -615,10 +622,10 @@ void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
   // npq quad (even bits only — 8-element fragment reads stay intact).
   auto issue_all = [&](int kk0, int b) {
     #pragma unroll
-    for (int t = 0; t < TA; ++t) {
+    for (int t = 0; t < TA * CH; ++t) {
       const int bid = t * 256 + tid;
       const int c4 = bid & (TM / 4 - 1);
-      const int r4 = bid / (TM / 4);
+      const int r4 = bid / (TM / 4);    // 0 .. 16*CH-1
       const int npq0 = kk0 + r4 * 4;
       const int mcol = m0 + c4 * 4;
       if (npq0 + 4 <= d.K && mcol + 4 <= d.M) {
@@ -639,7 +646,7 @@ void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
       }
     }
     #pragma unroll
-    for (int t = 0; t < TB; ++t) {
+    for (int t = 0; t < TB * CH; ++t) {
       const int bid = t * 256 + tid;
       const int c4 = bid & (TN / 4 - 1);
       const int r4 = bid / (TN / 4);
@@ -665,9 +672,11 @@ void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
       }
     }
   };
+  // swizzle note: q ranges over 16*CH quads; the XOR touches only bits 1-3
+  // (even, <16), so swizzled quads stay inside their row for any CH
   auto write_all = [&]() {
     #pragma unroll
-    for (int t = 0; t < TA; ++t) {
+    for (int t = 0; t < TA * CH; ++t) {
       const int bid = t * 256 + tid;
       const int c4 = bid & (TM / 4 - 1);
       const int r4 = bid / (TM / 4);
@@ -676,10 +685,10 @@ void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
       const int q = r4 ^ (c4 & 14);
       #pragma unroll
       for (int i = 0; i < 4; ++i)
-        *(uint2*)(sA + (c4 * 4 + i) * LDK + q * 4) = out[i];
+        *(uint2*)(sA + (c4 * 4 + i) * SLDK + q * 4) = out[i];
     }
     #pragma unroll
-    for (int t = 0; t < TB; ++t) {
+    for (int t = 0; t < TB * CH; ++t) {
       const int bid = t * 256 + tid;
       const int c4 = bid & (TN / 4 - 1);
       const int r4 = bid / (TN / 4);
@@ -688,24 +697,24 @@ void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
       const int q = r4 ^ (c4 & 14);
       #pragma unroll
       for (int i = 0; i < 4; ++i)
-        *(uint2*)(sB + (c4 * 4 + i) * LDK + q * 4) = out[i];
+        *(uint2*)(sB + (c4 * 4 + i) * SLDK + q * 4) = out[i];
     }
   };
   auto mfma_all = [&]() {
     #pragma unroll
-    for (int ks = 0; ks < BK; ks += 32) {
+    for (int ks = 0; ks < SK; ks += 32) {
       bf16x8 af[4], bf[4];
       #pragma unroll
       for (int mi = 0; mi < 4; ++mi) {
         const int row = wm + mi * 16 + fr;
         const int q = ((ks + fk) >> 2) ^ ((row >> 2) & 14);
-        af[mi] = *(const bf16x8*)&sA[row * LDK + q * 4];
+        af[mi] = *(const bf16x8*)&sA[row * SLDK + q * 4];
       }
       #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
         const int row = wn + ni * 16 + fr;
         const int q = ((ks + fk) >> 2) ^ ((row >> 2) & 14);
-        bf[ni] = *(const bf16x8*)&sB[row * LDK + q * 4];
+        bf[ni] = *(const bf16x8*)&sB[row * SLDK + q * 4];
       }
       #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
@@ -721,20 +730,20 @@ void conv_wgrad_v2_kernel(const __bf16* __restrict__ dy,
   //   meta(k+2) between two barriers
   bool first = true;
   int buf = 0;
-  for (int kc = blockIdx.z; kc < nchunks; kc += gridDim.z) {
+  for (int kc = blockIdx.z; kc < nstages; kc += gridDim.z) {
     const int kn = kc + gridDim.z;
     if (first) {
       compute_meta(kc, 0);
       __syncthreads();
-      issue_all(kc * BK, 0);
+      issue_all(kc * SK, 0);
       write_all();
       compute_meta(kn, 1);
       __syncthreads();
       first = false;
-      buf = 1;                          // meta[1] holds chunk kn
+      buf = 1;                          // meta[1] holds stage kn
     }
-    const bool has_next = kn < nchunks;
-    if (has_next) issue_all(kn * BK, buf);
+    const bool has_next = kn < nstages;
+    if (has_next) issue_all(kn * SK, buf);
     mfma_all();
     if (has_next) {
       __syncthreads();
@@ -1228,13 +1237,16 @@ at::Tensor conv_wgrad_igemm(at::Tensor dy, at::Tensor x, long R, long S,
   }
   if ((tile == 5 || tile == 6) && !lin_ok) tile = tile - 3;  // 2 / 3
   if (!fast || stem) tile = (tile == 4) ? 4 : 1;  // v2 needs the FAST layout
-  const int TM = (tile == 3 || tile == 6) ? 256
-               : (tile == 2 || tile == 5) ? 128 : 64;
-  const int TN = (tile == 2 || tile == 5) ? 128
-               : (tile == 3 || tile == 6) ? 64 : 128;
+  const int TM = (tile == 3 || tile == 6 || tile == 9) ? 256
+               : (tile == 2 || tile == 5 || tile == 8) ? 128 : 64;
+  const int TN = (tile == 2 || tile == 5 || tile == 8) ? 128
+               : (tile == 3 || tile == 6 || tile == 9) ? 64 : 128;
 
   const int tm = (d.M + TM - 1) / TM, tn = (d.N + TN - 1) / TN;
   const int nchunks = (d.K + BK - 1) / BK;
+  // wide-stage v2 (wtile 8/9) strides 128-row stages: a z-slice beyond the
+  // stage count would leave its partial slab unwritten
+  const int nzmax = (tile == 8 || tile == 9) ? (nchunks + 1) / 2 : nchunks;
   // heuristic: ~1024 blocks fills the chip, but keep >=8 chunks per block
   // (fewer and the per-block fill/drain + slab traffic dominates — measured
   // on the CIFAR shapes, where over-splitting cost 2-4x)
@@ -1242,13 +1254,15 @@ at::Tensor conv_wgrad_igemm(at::Tensor dy, at::Tensor x, long R, long S,
                                 : std::min(1024 / (tm * tn), nchunks / 8);
   // bound the partial-slab workspace to ~96 MB
   const long max_ws = 96L * 1024 * 1024 / ((long)d.M * d.N * 4);
-  splits = std::max(1, (int)std::min({(long)splits, (long)nchunks,
+  splits = std::max(1, (int)std::min({(long)splits, (long)nzmax,
                                       std::max(max_ws, 1L), 1024L}));
   const dim3 grid(tn, tm, splits);
 
   if (splits == 1 && !stem && tile != 4) {
     auto* kern =
-        tile == 6 ? conv_wgrad_lin_kernel<4, 1, true>
+        tile == 9 ? conv_wgrad_v2_kernel<4, 1, true, 2>
+        : tile == 8 ? conv_wgrad_v2_kernel<2, 2, true, 2>
+        : tile == 6 ? conv_wgrad_lin_kernel<4, 1, true>
         : tile == 5 ? conv_wgrad_lin_kernel<2, 2, true>
         : tile == 3 ? conv_wgrad_v2_kernel<4, 1, true>
         : tile == 2 ? conv_wgrad_v2_kernel<2, 2, true>
@@ -1265,7 +1279,9 @@ at::Tensor conv_wgrad_igemm(at::Tensor dy, at::Tensor x, long R, long S,
       ? at::zeros({MN}, x.options().dtype(at::kFloat))
       : at::empty({(long)splits * MN}, x.options().dtype(at::kFloat));
   auto* kern =
-      tile == 6 ? conv_wgrad_lin_kernel<4, 1, false>
+      tile == 9 ? conv_wgrad_v2_kernel<4, 1, false, 2>
+      : tile == 8 ? conv_wgrad_v2_kernel<2, 2, false, 2>
+      : tile == 6 ? conv_wgrad_lin_kernel<4, 1, false>
       : tile == 5 ? conv_wgrad_lin_kernel<2, 2, false>
       : tile == 3 ? conv_wgrad_v2_kernel<4, 1, false>
       : tile == 2 ? conv_wgrad_v2_kernel<2, 2, false>
