@@ -109,7 +109,7 @@ def test_conv_wgrad_splits_agree(splits):
         assert _rel_err(got, ref.float()) < 5e-3
 
 
-@pytest.mark.parametrize("wtile", [1, 2, 3, 4, 5, 6])
+@pytest.mark.parametrize("wtile", [1, 2, 3, 4, 5, 6, 8, 9])
 def test_conv_wgrad_tile_variants_agree(wtile):
     """v2's 128x128 / 256x64 tiles (register 4x4 transpose + ds_write_b64
     staging) compute the same dw as the v1 64x128 tile."""
