@@ -96,3 +96,13 @@ def test_resume_from_checkpoint(tmp_path, free_port):
     assert r.returncode == 0, r.stderr[-2000:]
     assert "resumed from" in r.stdout
     assert "Epoch: [1][0/" in r.stdout  # continued at epoch 1, not 0
+
+
+def test_apex_entry_o2_mode(tmp_path, free_port):
+    """distributed_apex.py --amp bf16_o2: the apex-O2-equivalent path runs
+    end to end (bf16 model, fp32 masters, train + eval + checkpoint)."""
+    r = _run(tmp_path, "distributed_apex.py",
+             ["--amp", "bf16_o2", "--port", str(free_port)])
+    assert "Epoch: [0][0/" in r.stdout
+    assert "Acc@1" in r.stdout
+    assert (tmp_path / "ckpts").exists()
