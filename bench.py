@@ -234,6 +234,10 @@ def main():
             },
         }
         print(json.dumps(result), flush=True)
+        if os.environ.get("MI355X_TUNE_REPORT", "0") == "1":
+            from mi355x_ddp.ops.conv import tune_report
+            print("[autotune choices]\n" + tune_report(), file=sys.stderr,
+                  flush=True)
     if world > 1:
         dist.destroy_process_group()
 

@@ -116,6 +116,17 @@ def clear_autotune_cache() -> None:
     _TUNE_CACHE.clear()
 
 
+def tune_report() -> str:
+    """Human-readable dump of the per-shape choices the tuner made
+    (``MI355X_TUNE_REPORT=1`` makes bench.py print it to stderr at exit)."""
+    lines = []
+    for key in sorted(_TUNE_CACHE, key=str):
+        choice = _TUNE_CACHE[key]
+        label = "MIOpen" if choice == MIOPEN else str(choice)
+        lines.append(f"{key[0]:>12} {str(key[1:]):<70} -> {label}")
+    return "\n".join(lines) if lines else "(autotune cache empty)"
+
+
 MIOPEN = -1  # tuner sentinel: this pass is fastest on the MIOpen kernel
 
 
