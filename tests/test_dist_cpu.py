@@ -387,3 +387,74 @@ def test_world1_fast_path_matches_static():
         results[mode] = [p.detach().clone() for p in net.parameters()]
     for a, b in zip(results["fast"], results["static"]):
         assert torch.allclose(a, b, atol=1e-7), (a - b).abs().max()
+
+
+def _check_grad_accu_with_bf16_comm(rank, world):
+    """no_sync micro-steps + bf16 comm buckets: accumulation stays fp32 and
+    the final synced step matches the fp32-comm path within bf16 wire
+    rounding."""
+    from mi355x_ddp.parallel import FlatDDP
+    results = {}
+    for comm in ("bf16", "fp32"):
+        torch.manual_seed(21)
+        model = nn.Sequential(nn.Linear(6, 12), nn.ReLU(), nn.Linear(12, 3))
+        wrapped = FlatDDP(model, bucket_cap_mb=1e-5,
+                          comm_dtype=torch.bfloat16 if comm == "bf16" else None)
+        gen = torch.Generator().manual_seed(4)
+        xs = [torch.randn(4, 6, generator=gen) for _ in range(2 * world)]
+        wrapped.zero_grad_buffer()
+        with wrapped.no_sync():
+            wrapped(xs[rank]).pow(2).mean().backward()
+            wrapped.finalize_backward()
+        wrapped(xs[world + rank]).pow(2).mean().backward()
+        wrapped.finalize_backward()
+        results[comm] = wrapped.flat_grads.clone()
+    assert torch.allclose(results["bf16"], results["fp32"],
+                          atol=5e-3, rtol=5e-2), \
+        (results["bf16"] - results["fp32"]).abs().max()
+
+
+def test_grad_accu_with_bf16_comm(free_port):
+    _run(_check_grad_accu_with_bf16_comm, free_port)
+
+
+def _check_syncbn_with_o2(rank, world):
+    """SyncBN with bf16 (O2) parameters: packed-stat all_reduce and the
+    backward still match the gathered-batch fp32 reference within bf16
+    tolerance, and ranks agree bitwise after the grad all-reduce."""
+    from mi355x_ddp.core.amp import cast_model_bf16
+    from mi355x_ddp.ops.batchnorm import MI355SyncBatchNorm
+    from mi355x_ddp.parallel import FlatDDP
+    torch.manual_seed(31)
+    net = nn.Sequential(nn.Conv2d(3, 8, 3, padding=1, bias=False),
+                        nn.BatchNorm2d(8))
+    net = MI355SyncBatchNorm.convert_sync_batchnorm(net)
+    net = cast_model_bf16(net)
+    wrapped = FlatDDP(net, grad_dtype=torch.bfloat16)
+    assert wrapped.flat_grads.dtype == torch.bfloat16
+    gen = torch.Generator().manual_seed(17)
+    xs = [torch.randn(2, 3, 8, 8, generator=gen).bfloat16()
+          for _ in range(world)]
+    wrapped.zero_grad_buffer()
+    out = wrapped(xs[rank])
+    out.float().pow(2).mean().backward()
+    wrapped.finalize_backward()
+    # ranks agree bitwise after the (bf16) gradient all-reduce
+    flat = wrapped.flat_grads.cpu()
+    gathered = [torch.zeros_like(flat) for _ in range(world)]
+    dist.all_gather(gathered, flat)
+    assert torch.equal(gathered[0], gathered[1])
+    # running stats were updated and stay finite with bf16 parameters;
+    # both ranks hold the SAME stats (they came from the packed all_reduce)
+    bn = net[1]
+    stats = torch.cat([bn.running_mean.float(), bn.running_var.float()]).cpu()
+    assert torch.isfinite(stats).all()
+    gathered_s = [torch.zeros_like(stats) for _ in range(world)]
+    dist.all_gather(gathered_s, stats)
+    assert torch.allclose(gathered_s[0], gathered_s[1], atol=1e-6)
+    assert not torch.equal(bn.running_mean.float(),
+                           torch.zeros_like(bn.running_mean.float()))
+
+
+def test_syncbn_with_o2(free_port):
+    _run(_check_syncbn_with_o2, free_port)
