@@ -33,6 +33,11 @@ class TrainConfig:
     weight_decay: float = 1e-4
     lr_milestones: List[int] = field(default_factory=lambda: [60, 120, 160])
     lr_gamma: float = 0.2
+    # scheduler: "multistep" (reference default, distributed.py:64) or
+    # "cosine"; warmup_epochs > 0 prepends a linear LR warmup (beyond the
+    # reference, standard large-batch practice)
+    lr_schedule: str = "multistep"
+    warmup_epochs: int = 0
     grad_accu_steps: int = 1          # micro-batching; collectives elided on non-final micro-steps
 
     # precision: "fp32" | "bf16" | "fp16" | "bf16_o2"
@@ -74,6 +79,7 @@ class TrainConfig:
     save_epoch: int = 15
     eval_every_epoch: bool = True
     resume: str = ""
+    evaluate: bool = False            # validation only (with --resume)
     # step caps (None = full epoch); used by smoke tests and quick CLI runs
     max_train_steps: Optional[int] = None
     max_eval_steps: Optional[int] = None
@@ -115,6 +121,11 @@ def add_common_args(p: argparse.ArgumentParser) -> argparse.ArgumentParser:
     p.add_argument("--max_eval_steps", default=None, type=int)
     p.add_argument("--dist_timeout_s", default=None, type=int,
                    help="process-group collective timeout in seconds")
+    p.add_argument("--lr_schedule", default=None, type=str,
+                   choices=[None, "multistep", "cosine"])
+    p.add_argument("--warmup_epochs", default=None, type=int)
+    p.add_argument("--evaluate", action="store_true",
+                   help="run validation only (use with --resume)")
     return p
 
 
@@ -144,5 +155,11 @@ def config_from_args(args: argparse.Namespace, **overrides) -> TrainConfig:
         kw["max_eval_steps"] = args.max_eval_steps
     if getattr(args, "dist_timeout_s", None) is not None:
         kw["dist_timeout_s"] = args.dist_timeout_s
+    if getattr(args, "lr_schedule", None):
+        kw["lr_schedule"] = args.lr_schedule
+    if getattr(args, "warmup_epochs", None) is not None:
+        kw["warmup_epochs"] = args.warmup_epochs
+    if getattr(args, "evaluate", False):
+        kw["evaluate"] = True
     kw.update(overrides)
     return TrainConfig(**kw)

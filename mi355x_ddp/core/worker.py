@@ -83,10 +83,28 @@ def build_training(cfg: TrainConfig, device: torch.device, world_size: int,
     criterion = SoftmaxCrossEntropy().to(device)
     optimizer = FusedSGD(model.parameters(), lr=cfg.lr, momentum=cfg.momentum,
                          weight_decay=cfg.weight_decay)
-    scheduler = torch.optim.lr_scheduler.MultiStepLR(
-        optimizer, milestones=cfg.lr_milestones, gamma=cfg.lr_gamma)
+    scheduler = build_scheduler(optimizer, cfg)
     scaler = build_scaler(cfg.amp)
     return model, criterion, optimizer, scheduler, scaler
+
+
+def build_scheduler(optimizer, cfg: TrainConfig):
+    """MultiStepLR (the reference's schedule, distributed.py:64) or cosine,
+    optionally behind a linear warmup."""
+    sched = getattr(cfg, "lr_schedule", "multistep")
+    warm = max(0, getattr(cfg, "warmup_epochs", 0))
+    if sched == "cosine":
+        main = torch.optim.lr_scheduler.CosineAnnealingLR(
+            optimizer, T_max=max(1, cfg.epochs - warm))
+    else:
+        main = torch.optim.lr_scheduler.MultiStepLR(
+            optimizer, milestones=cfg.lr_milestones, gamma=cfg.lr_gamma)
+    if warm == 0:
+        return main
+    warmup = torch.optim.lr_scheduler.LinearLR(
+        optimizer, start_factor=1.0 / (warm + 1), total_iters=warm)
+    return torch.optim.lr_scheduler.SequentialLR(
+        optimizer, [warmup, main], milestones=[warm])
 
 
 def main_worker(local_rank: int, nprocs: int, cfg: TrainConfig,
@@ -111,6 +129,11 @@ def main_worker(local_rank: int, nprocs: int, cfg: TrainConfig,
             print(f"resumed from {cfg.resume} at epoch {start_epoch}")
     train_loader, test_loader, train_sampler = build_loaders(
         cfg, nprocs, rank, distributed=nprocs > 1)
+    if cfg.evaluate:
+        from .engine import validate
+        best = validate(model, test_loader, criterion, device, cfg)
+        dist_utils.cleanup()
+        return best
     best = fit(model, train_loader, test_loader, train_sampler, criterion,
                optimizer, scheduler, cfg, device, scaler=scaler,
                start_epoch=start_epoch)

@@ -42,3 +42,42 @@ def test_grad_accu_flag():
     add_common_args(p)
     cfg = config_from_args(p.parse_args(["--grad_accu_steps", "4"]))
     assert cfg.grad_accu_steps == 4
+
+
+def test_scheduler_options_and_warmup():
+    """cosine + warmup: LR ramps linearly for warmup_epochs then follows
+    the cosine; multistep default matches the reference schedule."""
+    import torch
+    from mi355x_ddp.config import TrainConfig
+    from mi355x_ddp.core.worker import build_scheduler
+    from mi355x_ddp.ops import FusedSGD
+
+    p = torch.nn.Parameter(torch.zeros(1))
+    cfg = TrainConfig(lr=1.0, epochs=10, lr_schedule="cosine", warmup_epochs=2)
+    opt = FusedSGD([p], lr=cfg.lr)
+    sched = build_scheduler(opt, cfg)
+    lrs = []
+    for _ in range(10):
+        lrs.append(opt.param_groups[0]["lr"])
+        opt.step()
+        sched.step()
+    assert lrs[0] < lrs[1] < lrs[2]          # warmup ramps up
+    assert abs(lrs[2] - 1.0) < 1e-6          # reaches base lr
+    assert lrs[-1] < lrs[3]                  # cosine decays after warmup
+
+    cfg2 = TrainConfig(lr=1.0, epochs=200)   # reference default schedule
+    opt2 = FusedSGD([torch.nn.Parameter(torch.zeros(1))], lr=1.0)
+    s2 = build_scheduler(opt2, cfg2)
+    assert isinstance(s2, torch.optim.lr_scheduler.MultiStepLR)
+
+
+def test_evaluate_only_mode(tmp_path):
+    """--evaluate runs validation without training (reference had no such
+    mode; standard for checkpoint checking)."""
+    from mi355x_ddp.config import TrainConfig
+    from mi355x_ddp.core.worker import main_worker
+    cfg = TrainConfig(batch_size=16, num_workers=0, synthetic=True,
+                      evaluate=True, max_eval_steps=2, sync_bn=False,
+                      metrics_dir=str(tmp_path), ckpt_dir=str(tmp_path))
+    acc = main_worker(0, 1, cfg, init_pg=False)
+    assert 0.0 <= acc <= 100.0
