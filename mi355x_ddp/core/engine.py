@@ -197,7 +197,12 @@ def fit(model, train_loader, test_loader, train_sampler, criterion, optimizer,
         if cfg.eval_every_epoch:
             acc = validate(model, test_loader, criterion, device, cfg,
                            sink=sink, epoch=epoch)
-            best_acc = max(best_acc, acc)
+            if acc > best_acc:
+                best_acc = acc
+                # keep the best checkpoint alongside the periodic ones
+                save_checkpoint(cfg.ckpt_dir, cfg.arch, epoch, model,
+                                optimizer, scheduler, scaler, best_acc,
+                                tag="best")
         if cfg.save_epoch > 0 and (epoch + 1) % cfg.save_epoch == 0:
             save_checkpoint(cfg.ckpt_dir, cfg.arch, epoch, model, optimizer,
                             scheduler, scaler, best_acc)

@@ -218,3 +218,19 @@ def test_watchdog_none_is_noop():
     from mi355x_ddp.core.watchdog import StepWatchdog
     with StepWatchdog(None):
         pass  # must not arm faulthandler at all
+
+
+def test_best_checkpoint_saved(tmp_path):
+    """fit() persists {arch}_best.pt when validation improves."""
+    init_seeds(0)
+    cfg = _tiny_cfg(ckpt_dir=str(tmp_path), metrics_dir=str(tmp_path),
+                    max_train_steps=2, max_eval_steps=2, save_epoch=0)
+    device = torch.device("cpu")
+    model, crit, opt, sched, scaler = build_training(
+        cfg, device, 1, 0, distributed=True, wrap="flat")
+    from mi355x_ddp.core.engine import fit
+    from mi355x_ddp.data import build_loaders
+    tl, vl, _ = build_loaders(cfg, 1, 0, distributed=False)
+    fit(model, tl, vl, None, crit, opt, sched, cfg, device, scaler=scaler)
+    assert (tmp_path / "resnet18_best.pt").exists()
+    assert (tmp_path / "resnet18_final.pt").exists()
