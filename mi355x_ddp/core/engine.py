@@ -197,7 +197,9 @@ def fit(model, train_loader, test_loader, train_sampler, criterion, optimizer,
         if cfg.eval_every_epoch:
             acc = validate(model, test_loader, criterion, device, cfg,
                            sink=sink, epoch=epoch)
-            if acc > best_acc:
+            # >= so the first evaluation (possibly 0.0 on tiny synthetic
+            # runs) still produces a best checkpoint
+            if acc >= best_acc:
                 best_acc = acc
                 # keep the best checkpoint alongside the periodic ones
                 save_checkpoint(cfg.ckpt_dir, cfg.arch, epoch, model,
