@@ -27,6 +27,9 @@ def main():
     cfg = config_from_args(args)
     if cfg.amp == "fp32":  # default for this entry is mixed precision
         cfg = cfg.replace(amp="bf16")
+    if getattr(args, "channels_last", None) is None and             torch.cuda.is_available():
+        # the AMP entry defaults to the fast path: NHWC + native igemm conv
+        cfg = cfg.replace(channels_last=True)
     nprocs = args.nprocs or torch.cuda.device_count() or 1
     if nprocs == 1:
         main_worker(0, 1, cfg)

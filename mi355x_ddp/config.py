@@ -123,6 +123,11 @@ def add_common_args(p: argparse.ArgumentParser) -> argparse.ArgumentParser:
                    help="process-group collective timeout in seconds")
     p.add_argument("--lr_schedule", default=None, type=str,
                    choices=[None, "multistep", "cosine"])
+    p.add_argument("--channels-last", "--channels_last", dest="channels_last",
+                   action="store_true", default=None,
+                   help="NHWC layout + native implicit-GEMM conv kernels")
+    p.add_argument("--no-channels-last", dest="channels_last",
+                   action="store_false")
     p.add_argument("--warmup_epochs", default=None, type=int)
     p.add_argument("--evaluate", action="store_true",
                    help="run validation only (use with --resume)")
@@ -161,5 +166,7 @@ def config_from_args(args: argparse.Namespace, **overrides) -> TrainConfig:
         kw["warmup_epochs"] = args.warmup_epochs
     if getattr(args, "evaluate", False):
         kw["evaluate"] = True
+    if getattr(args, "channels_last", None) is not None:
+        kw["channels_last"] = bool(args.channels_last)
     kw.update(overrides)
     return TrainConfig(**kw)
